@@ -1,0 +1,108 @@
+"""Raw (headerless) image I/O: whole-frame and per-rank block access.
+
+Mirrors the reference's raw-image contract (BASELINE.json:5: "raw
+input/output image format"; SURVEY.md §2.1 R6: each rank reads/writes its
+own block at computed offsets). Host-side I/O sits OUTSIDE the timed
+iteration loop, exactly as the reference times only the MPI loop
+(SURVEY.md §5.8c).
+
+A GPU-box path with the same block-offset semantics also exists natively in
+C++ (csrc/io.cpp, `_C.read_block`/`_C.write_block`) for the engine's
+scatter/gather; this module is the portable implementation and the oracle
+for it.
+"""
+
+from __future__ import annotations
+
+import os
+
+import numpy as np
+
+from ..spec import DType
+
+
+def read_image(path: str, height: int, width: int, channels: int, dtype: DType) -> np.ndarray:
+    """Read a whole raw frame as (height, width*channels) of dtype."""
+    n = height * width * channels
+    arr = np.fromfile(path, dtype=dtype.numpy_dtype, count=n)
+    if arr.size != n:
+        raise ValueError(
+            f"{path}: expected {n} elements ({height}x{width}x{channels}), got {arr.size}"
+        )
+    return arr.reshape(height, width * channels)
+
+
+def write_image(path: str, img: np.ndarray) -> None:
+    np.ascontiguousarray(img).tofile(path)
+
+
+def read_block(
+    path: str,
+    height: int,
+    width: int,
+    channels: int,
+    dtype: DType,
+    y0: int,
+    x0: int,
+    h: int,
+    w: int,
+) -> np.ndarray:
+    """Read one rank's (h, w) pixel block: returns (h, w*channels).
+
+    Seeks to each row's offset like the reference's per-rank fread+fseek
+    (SURVEY.md §2.3 'input scatter' row).
+    """
+    row_elems = width * channels
+    itemsize = dtype.itemsize
+    out = np.empty((h, w * channels), dtype=dtype.numpy_dtype)
+    with open(path, "rb") as f:
+        for r in range(h):
+            f.seek(((y0 + r) * row_elems + x0 * channels) * itemsize)
+            buf = f.read(w * channels * itemsize)
+            if len(buf) != w * channels * itemsize:
+                raise ValueError(f"{path}: short read at row {y0+r}")
+            out[r] = np.frombuffer(buf, dtype=dtype.numpy_dtype)
+    return out
+
+
+def write_block(
+    path: str,
+    height: int,
+    width: int,
+    channels: int,
+    dtype: DType,
+    y0: int,
+    x0: int,
+    block: np.ndarray,
+) -> None:
+    """Write one rank's block into a (pre-sized) raw file at its offsets."""
+    h = block.shape[0]
+    w_elems = block.shape[1]
+    row_elems = width * channels
+    itemsize = dtype.itemsize
+    total = height * row_elems * itemsize
+    # Ensure the file exists at full size (rank-parallel writers each own
+    # disjoint regions; os.O_CREAT + pwrite keeps them independent).
+    fd = os.open(path, os.O_RDWR | os.O_CREAT, 0o644)
+    try:
+        if os.fstat(fd).st_size < total:
+            os.ftruncate(fd, total)
+        blk = np.ascontiguousarray(block)
+        for r in range(h):
+            off = ((y0 + r) * row_elems + x0 * channels) * itemsize
+            os.pwrite(fd, blk[r].tobytes(), off)
+    finally:
+        os.close(fd)
+
+
+def synthetic_image(
+    height: int, width: int, channels: int, dtype: DType, seed: int = 1234
+) -> np.ndarray:
+    """Seeded random-pixel frame (BASELINE.json north-star: benchmarks run
+    'on synthetic random-pixel images'); (height, width*channels)."""
+    rng = np.random.default_rng(seed)
+    shape = (height, width * channels)
+    if dtype == DType.U8:
+        return rng.integers(0, 256, size=shape, dtype=np.uint8)
+    # float images in [0, 256) — same dynamic range as u8
+    return (rng.random(shape, dtype=np.float32) * 256).astype(dtype.numpy_dtype)
