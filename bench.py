@@ -1,0 +1,159 @@
+"""Flagship benchmark — the driver contract.
+
+    python bench.py --gpus N --steps K --warmup W
+
+One step = one conv iteration of the headline config (BASELINE.json:2):
+16384² grayscale u8, 3×3 Gaussian blur, spatially decomposed over N GPUs
+(2×4 grid at 8) with RCCL halo exchange over xGMI. Prints one JSON line
+from rank 0; `value` = whole-job Gpixel-iters/sec (image pixels × steps /
+max-over-ranks elapsed). Strong scaling: the image is fixed as N grows.
+
+For N>1 the driver launches this under torch.distributed.run (one rank
+per GPU); RANK/LOCAL_RANK/WORLD_SIZE come from the env.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import numpy as np  # noqa: E402
+import torch  # noqa: E402
+import torch.distributed as dist  # noqa: E402
+
+from parallel_convolution_amd.models.filters import (  # noqa: E402
+    GAUSSIAN_BLUR_3,
+    SHARPEN_5,
+)
+from parallel_convolution_amd.parallel import bootstrap  # noqa: E402
+from parallel_convolution_amd.parallel.decomp import make_blocks  # noqa: E402
+from parallel_convolution_amd.spec import DType, ProblemSpec  # noqa: E402
+from parallel_convolution_amd.utils.rawio import synthetic_image  # noqa: E402
+
+PRESETS = {
+    # BASELINE.json:6-12 configs 2-5 (config 1 is the CPU plumbing test)
+    "headline": dict(width=16384, height=16384, channels=1, dtype="u8", filt="blur3"),
+    "rgb5": dict(width=16384, height=16384, channels=3, dtype="u8", filt="sharpen5"),
+    "fp16-65536": dict(width=65536, height=65536, channels=1, dtype="fp16", filt="blur3"),
+    "single-8192": dict(width=8192, height=8192, channels=1, dtype="u8", filt="blur3"),
+}
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=100)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--preset", default="headline", choices=sorted(PRESETS))
+    ap.add_argument("--chunk", type=int, default=None, help="halo-deepening chunk T")
+    ap.add_argument("--json-out", default=None)
+    args = ap.parse_args()
+
+    world = bootstrap.env_world()
+    rank = bootstrap.env_rank()
+    local_rank = bootstrap.env_local_rank()
+    n_gpus = max(args.gpus, world)
+
+    if world != n_gpus and world > 1:
+        raise SystemExit(f"WORLD_SIZE={world} but --gpus={args.gpus}")
+
+    p = PRESETS[args.preset]
+    spec = ProblemSpec(
+        p["width"],
+        p["height"],
+        p["channels"],
+        DType(p["dtype"]),
+        GAUSSIAN_BLUR_3 if p["filt"] == "blur3" else SHARPEN_5,
+        args.steps,
+    )
+
+    if world > 1:
+        bootstrap.init_control_group("gloo")
+    torch.cuda.set_device(local_rank)
+
+    from parallel_convolution_amd.hip_engine import HipEngine
+
+    blocks = make_blocks(spec.height, spec.width, n_gpus, spec.channels)
+    if world > 1:
+        my_blocks = [blocks[rank]]
+    else:
+        # single-process N-block mode is only used for n_gpus==1
+        assert n_gpus == 1, "multi-GPU runs must be launched via torchrun"
+        my_blocks = blocks
+
+    eng = HipEngine(spec, my_blocks, world_blocks=blocks, chunk_iters=args.chunk)
+    if world > 1:
+        eng.init_distributed(rank, world)
+
+    # synthetic random-pixel block data (BASELINE north-star), per-rank seed
+    for i, bs in enumerate(my_blocks):
+        eng.load_block(i, synthetic_image(bs.h, bs.w, spec.channels, spec.dtype, seed=1234 + bs.rank))
+
+    # warmup (untimed)
+    if args.warmup > 0:
+        eng.run(args.warmup)
+    torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    torch.cuda.synchronize()
+
+    t0 = time.perf_counter()
+    eng.run(args.steps)
+    torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    elapsed = t1 - t0
+
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+        dist.barrier()
+
+    pixel_iters = float(spec.width) * spec.height * args.steps
+    value = pixel_iters / elapsed / 1e9  # Gpixel-iters/s, whole job
+
+    if rank == 0:
+        result = {
+            "metric": "Gpixel-iters/sec",
+            "value": round(value, 3),
+            "unit": "Gpixel-iters/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed * 1000.0 / args.steps, 6),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": spec.dtype.value,
+            "data": "synthetic random-pixel image, seeded per rank",
+            "config": {
+                "model": f"iterconv-{spec.k}x{spec.k}-"
+                f"{'gray' if spec.channels == 1 else 'rgb'}-{spec.dtype.value}",
+                "image": f"{spec.width}x{spec.height}x{spec.channels}",
+                "filter": p["filt"],
+                "accumulate": "fp32",
+                "global_batch": 1,
+                "seq_len": None,
+                "parallelism": "spatial2d-{}x{}".format(*blocks[0].grid),
+                "halo_chunk_iters": eng.chunk_iters,
+                "preset": args.preset,
+            },
+        }
+        line = json.dumps(result)
+        print(line, flush=True)
+        if args.json_out:
+            with open(args.json_out, "a") as f:
+                f.write(line + "\n")
+
+    if world > 1:
+        bootstrap.finalize()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

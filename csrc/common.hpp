@@ -1,0 +1,69 @@
+// Shared declarations for the MI355X-native convolution engine.
+//
+// Built for gfx950 (CDNA4) only — no CUDA compatibility paths
+// (BASELINE.json north-star: "no Triton, no CUDA-compat shims, no dual
+// code paths").
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <cstdio>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#define HIP_CHECK(expr)                                                        \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) {                                                    \
+      throw std::runtime_error(std::string("HIP error: ") +                    \
+                               hipGetErrorString(_e) + " at " __FILE__ ":" +   \
+                               std::to_string(__LINE__));                      \
+    }                                                                          \
+  } while (0)
+
+namespace pconv {
+
+enum class DType : int { U8 = 0, F16 = 1, F32 = 2 };
+
+inline int itemsize(DType d) { return d == DType::U8 ? 1 : d == DType::F16 ? 2 : 4; }
+
+// Geometry of one rank's padded block buffer. All x-quantities are in
+// storage ELEMENTS (pixel x * channels) unless suffixed Px (pixels).
+// The owned block occupies rows [haloPx, haloPx+h) and element columns
+// [haloPx*S, (haloPx+w)*S) of a (bufh, pitchE) buffer.
+struct BlockGeom {
+  int h = 0;        // owned rows
+  int wPx = 0;      // owned width in pixels
+  int haloPx = 0;   // allocated halo width in pixels (>= max exchange depth)
+  int S = 1;        // channels (element stride of one pixel)
+  int pitchE = 0;   // row pitch in elements
+  int bufh = 0;     // buffer rows = h + 2*haloPx
+  // neighbors: index meaning depends on engine mode (local block index or
+  // global NCCL rank); -1 = physical edge. Order: N, S, W, E.
+  int nbr[4] = {-1, -1, -1, -1};
+};
+
+enum { NB_N = 0, NB_S = 1, NB_W = 2, NB_E = 3 };
+
+// Compute region in buffer coords: rows [py0,py1), element cols [ex0,ex1).
+struct Region {
+  int py0, py1, ex0, ex1;
+};
+
+// Launch the k×k convolution over a region.
+//   src/dst: element pointers to the padded buffers
+//   wgt: k*k fp32 filter on device
+// Fast LDS-tiled path for radius<=2 and S in {1,3}; generic fallback
+// otherwise.
+void launch_conv(DType dt, int S, int k, const void* src, void* dst,
+                 const float* wgt, const BlockGeom& g, const Region& r,
+                 hipStream_t stream);
+
+// max |dst - src| over the owned region, accumulated into *resultDev
+// (a single float, caller zeroes it first).
+void launch_diff_max(DType dt, const void* src, const void* dst,
+                     const BlockGeom& g, float* resultDev, hipStream_t stream);
+
+}  // namespace pconv

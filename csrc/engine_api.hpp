@@ -1,0 +1,313 @@
+// Per-GPU runtime: buffers, streams, chunked iterate/exchange loop.
+//
+// MI355X-native replacement for the reference's MPI main loop
+// (Irecv/Isend → convolute → Waitall → border convolute →
+// swap). Differences, by design (SURVEY.md §7 hard parts):
+//
+//  * Halo deepening: exchange depth d = r*T once per T iterations, then T
+//    kernel launches over shrinking regions — amortizes RCCL latency,
+//    which on xGMI dominates the KB-scale halo volume.
+//  * Two-phase corner-free exchange (parallel/decomp.py docstring):
+//    phase A = E/W column strips (strided → staged via hipMemcpy2DAsync),
+//    phase B = N/S row strips at full pitch (contiguous, sent in place).
+//  * Everything is enqueued asynchronously on two HIP streams (compute +
+//    comm) with event ordering; run() performs ZERO host synchronization
+//    until the end of the whole iteration loop.
+//
+// Communication backends:
+//  * LOCAL: all blocks of the decomposition live in this process on one
+//    GPU; "exchange" = device-to-device copies. This is the test mode that
+//    validates pack/geometry on a 1-GPU box.
+//  * RCCL: one block, neighbors are global ranks; grouped ncclSend/Recv
+//    over xGMI (SURVEY.md §2.3 mapping), communicator built from a
+//    Python-broadcast ncclUniqueId.
+
+#pragma once
+
+#include "common.hpp"
+
+#include <rccl/rccl.h>
+
+#include <algorithm>
+#include <array>
+#include <cstring>
+#include <memory>
+
+#define NCCL_CHECK(expr)                                                      \
+  do {                                                                        \
+    ncclResult_t _r = (expr);                                                 \
+    if (_r != ncclSuccess) {                                                  \
+      throw std::runtime_error(std::string("RCCL error: ") +                  \
+                               ncclGetErrorString(_r) + " at " __FILE__ ":" + \
+                               std::to_string(__LINE__));                     \
+    }                                                                         \
+  } while (0)
+
+namespace pconv {
+
+struct DeviceBlock {
+  BlockGeom g;
+  void* src = nullptr;  // swapped every iteration
+  void* dst = nullptr;
+  // staging for phase-A column strips (E/W): send+recv, each
+  // h * depth_max * S elements
+  void* sendW = nullptr;
+  void* sendE = nullptr;
+  void* recvW = nullptr;
+  void* recvE = nullptr;
+};
+
+class ConvEngine {
+ public:
+  ConvEngine(int dtype, int k, const std::vector<float>& filter, int channels)
+      : dt_(static_cast<DType>(dtype)), k_(k), S_(channels) {
+    if ((int)filter.size() != k * k) throw std::runtime_error("filter size != k*k");
+    HIP_CHECK(hipStreamCreateWithFlags(&compute_, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamCreateWithFlags(&comm_, hipStreamNonBlocking));
+    HIP_CHECK(hipEventCreateWithFlags(&evComm_, hipEventDisableTiming));
+    HIP_CHECK(hipEventCreateWithFlags(&evCompute_, hipEventDisableTiming));
+    HIP_CHECK(hipMalloc(&wgt_, filter.size() * sizeof(float)));
+    HIP_CHECK(hipMemcpy(wgt_, filter.data(), filter.size() * sizeof(float),
+                        hipMemcpyHostToDevice));
+    HIP_CHECK(hipMalloc(&diffDev_, sizeof(float)));
+  }
+
+  ~ConvEngine() {
+    for (auto& b : blocks_) {
+      if (b.sendW) (void)hipFree(b.sendW);
+      if (b.sendE) (void)hipFree(b.sendE);
+      if (b.recvW) (void)hipFree(b.recvW);
+      if (b.recvE) (void)hipFree(b.recvE);
+    }
+    if (wgt_) (void)hipFree(wgt_);
+    if (diffDev_) (void)hipFree(diffDev_);
+    if (comm_nccl_) (void)ncclCommDestroy(comm_nccl_);
+    (void)hipEventDestroy(evComm_);
+    (void)hipEventDestroy(evCompute_);
+    (void)hipStreamDestroy(compute_);
+    (void)hipStreamDestroy(comm_);
+  }
+
+  int add_block(uintptr_t src, uintptr_t dst, const BlockGeom& g) {
+    DeviceBlock b;
+    b.g = g;
+    b.src = reinterpret_cast<void*>(src);
+    b.dst = reinterpret_cast<void*>(dst);
+    const size_t strip = (size_t)g.h * g.haloPx * g.S * itemsize(dt_);
+    if (g.nbr[NB_W] >= 0) {
+      HIP_CHECK(hipMalloc(&b.sendW, strip));
+      HIP_CHECK(hipMalloc(&b.recvW, strip));
+    }
+    if (g.nbr[NB_E] >= 0) {
+      HIP_CHECK(hipMalloc(&b.sendE, strip));
+      HIP_CHECK(hipMalloc(&b.recvE, strip));
+    }
+    blocks_.push_back(b);
+    return (int)blocks_.size() - 1;
+  }
+
+  void init_nccl(const std::string& uid, int rank, int world) {
+    if (blocks_.size() != 1)
+      throw std::runtime_error("NCCL mode expects exactly one local block");
+    ncclUniqueId id;
+    if (uid.size() < sizeof(id)) throw std::runtime_error("bad nccl uid size");
+    std::memcpy(&id, uid.data(), sizeof(id));
+    NCCL_CHECK(ncclCommInitRank(&comm_nccl_, world, id, rank));
+    rank_ = rank;
+  }
+
+  static std::string get_unique_id() {
+    ncclUniqueId id;
+    NCCL_CHECK(ncclGetUniqueId(&id));
+    return std::string(reinterpret_cast<char*>(&id), sizeof(id));
+  }
+
+  // Run `iters` iterations in chunks of <= chunkT (exchange depth r*T per
+  // chunk). Fully asynchronous; blocks until the device finishes.
+  void run(int iters, int chunkT) {
+    const int r = k_ / 2;
+    int left = iters;
+    while (left > 0) {
+      const int T = std::min(chunkT, left);
+      const int d = r * T;
+      if (r > 0 && d > 0) exchange(d);
+      for (int t = 0; t < T; ++t) {
+        const int ext = d - (t + 1) * r;
+        for (auto& b : blocks_) {
+          launch_conv(dt_, S_, k_, b.src, b.dst, (const float*)wgt_, b.g,
+                      region_for(b.g, ext), compute_);
+        }
+        for (auto& b : blocks_) std::swap(b.src, b.dst);
+        swapped_ = !swapped_;
+      }
+      left -= T;
+    }
+    HIP_CHECK(hipStreamSynchronize(compute_));
+    HIP_CHECK(hipStreamSynchronize(comm_));
+    HIP_CHECK(hipGetLastError());
+  }
+
+  // max |src - dst| over owned regions of all local blocks (convergence
+  // test). Synchronous.
+  float diff_max() {
+    HIP_CHECK(hipMemsetAsync(diffDev_, 0, sizeof(float), compute_));
+    for (auto& b : blocks_)
+      launch_diff_max(dt_, b.src, b.dst, b.g, (float*)diffDev_, compute_);
+    float out = 0.f;
+    HIP_CHECK(hipMemcpyAsync(&out, diffDev_, sizeof(float),
+                             hipMemcpyDeviceToHost, compute_));
+    HIP_CHECK(hipStreamSynchronize(compute_));
+    return out;
+  }
+
+  // true if the latest result lives in the tensor originally passed
+  // as dst (odd total iteration count)
+  bool swapped() const { return swapped_; }
+
+  void synchronize() {
+    HIP_CHECK(hipStreamSynchronize(compute_));
+    HIP_CHECK(hipStreamSynchronize(comm_));
+  }
+
+ private:
+  Region region_for(const BlockGeom& g, int ext) const {
+    Region r;
+    r.py0 = g.haloPx - (g.nbr[NB_N] >= 0 ? ext : 0);
+    r.py1 = g.haloPx + g.h + (g.nbr[NB_S] >= 0 ? ext : 0);
+    const int pxLo = g.haloPx - (g.nbr[NB_W] >= 0 ? ext : 0);
+    const int pxHi = g.haloPx + g.wPx + (g.nbr[NB_E] >= 0 ? ext : 0);
+    r.ex0 = pxLo * g.S;
+    r.ex1 = pxHi * g.S;
+    return r;
+  }
+
+  // --- exchange ------------------------------------------------------------
+
+  char* elem(DeviceBlock& b, void* base, int row, int colPx) const {
+    return (char*)base +
+           ((size_t)row * b.g.pitchE + (size_t)colPx * b.g.S) * itemsize(dt_);
+  }
+
+  void copy2d(void* dst, size_t dpitch, const void* src, size_t spitch,
+              size_t widthB, size_t rows) {
+    HIP_CHECK(hipMemcpy2DAsync(dst, dpitch, src, spitch, widthB, rows,
+                               hipMemcpyDeviceToDevice, comm_));
+  }
+
+  void exchange(int d) {
+    // comm must see the latest src (written on compute)
+    HIP_CHECK(hipEventRecord(evCompute_, compute_));
+    HIP_CHECK(hipStreamWaitEvent(comm_, evCompute_, 0));
+    if (comm_nccl_) {
+      exchange_nccl(d);
+    } else {
+      exchange_local(d);
+    }
+    HIP_CHECK(hipEventRecord(evComm_, comm_));
+    HIP_CHECK(hipStreamWaitEvent(compute_, evComm_, 0));
+  }
+
+  // LOCAL mode: neighbors are local block indices; D2D copies.
+  void exchange_local(int d) {
+    const int is = itemsize(dt_);
+    // Phase A: E/W column strips (owned rows only)
+    for (auto& b : blocks_) {
+      if (b.g.nbr[NB_E] < 0) continue;
+      DeviceBlock& e = blocks_[b.g.nbr[NB_E]];
+      const int H = b.g.haloPx;
+      const size_t wB = (size_t)d * b.g.S * is;
+      const size_t bp = (size_t)b.g.pitchE * is;
+      const size_t ep = (size_t)e.g.pitchE * is;
+      // my east strip -> east's west halo
+      copy2d(elem(e, e.src, H, H - d), ep, elem(b, b.src, H, H + b.g.wPx - d),
+             bp, wB, b.g.h);
+      // east's west strip -> my east halo
+      copy2d(elem(b, b.src, H, H + b.g.wPx), bp, elem(e, e.src, H, H), ep, wB,
+             b.g.h);
+    }
+    // Phase B: N/S full-pitch row strips (contiguous)
+    for (auto& b : blocks_) {
+      if (b.g.nbr[NB_S] < 0) continue;
+      DeviceBlock& s = blocks_[b.g.nbr[NB_S]];
+      const int H = b.g.haloPx;
+      const size_t rowB = (size_t)b.g.pitchE * is;
+      // my bottom d rows -> south's north halo
+      HIP_CHECK(hipMemcpyAsync(elem(s, s.src, H - d, 0),
+                               elem(b, b.src, H + b.g.h - d, 0), rowB * d,
+                               hipMemcpyDeviceToDevice, comm_));
+      // south's top d rows -> my south halo
+      HIP_CHECK(hipMemcpyAsync(elem(b, b.src, H + b.g.h, 0),
+                               elem(s, s.src, H, 0), rowB * d,
+                               hipMemcpyDeviceToDevice, comm_));
+    }
+  }
+
+  // RCCL mode: one block, neighbors are global ranks.
+  void exchange_nccl(int d) {
+    DeviceBlock& b = blocks_[0];
+    const int is = itemsize(dt_);
+    const int H = b.g.haloPx;
+    const size_t stripB = (size_t)b.g.h * d * b.g.S * is;  // phase-A message
+    const size_t wB = (size_t)d * b.g.S * is;              // strip row bytes
+    const size_t bp = (size_t)b.g.pitchE * is;
+    const int W = b.g.nbr[NB_W], E = b.g.nbr[NB_E];
+    const int N = b.g.nbr[NB_N], Sn = b.g.nbr[NB_S];
+
+    // Phase A: pack E/W strips, grouped send/recv, unpack.
+    if (W >= 0)
+      copy2d(b.sendW, wB, elem(b, b.src, H, H), bp, wB, b.g.h);
+    if (E >= 0)
+      copy2d(b.sendE, wB, elem(b, b.src, H, H + b.g.wPx - d), bp, wB, b.g.h);
+    if (W >= 0 || E >= 0) {
+      NCCL_CHECK(ncclGroupStart());
+      if (W >= 0) {
+        NCCL_CHECK(ncclSend(b.sendW, stripB, ncclChar, W, comm_nccl_, comm_));
+        NCCL_CHECK(ncclRecv(b.recvW, stripB, ncclChar, W, comm_nccl_, comm_));
+      }
+      if (E >= 0) {
+        NCCL_CHECK(ncclSend(b.sendE, stripB, ncclChar, E, comm_nccl_, comm_));
+        NCCL_CHECK(ncclRecv(b.recvE, stripB, ncclChar, E, comm_nccl_, comm_));
+      }
+      NCCL_CHECK(ncclGroupEnd());
+      if (W >= 0)
+        copy2d(elem(b, b.src, H, H - d), bp, b.recvW, wB, wB, b.g.h);
+      if (E >= 0)
+        copy2d(elem(b, b.src, H, H + b.g.wPx), bp, b.recvE, wB, wB, b.g.h);
+    }
+
+    // Phase B: N/S full-pitch rows, contiguous — no staging.
+    if (N >= 0 || Sn >= 0) {
+      const size_t rowsB = (size_t)d * bp;
+      NCCL_CHECK(ncclGroupStart());
+      if (N >= 0) {
+        NCCL_CHECK(ncclSend(elem(b, b.src, H, 0), rowsB, ncclChar, N,
+                            comm_nccl_, comm_));
+        NCCL_CHECK(ncclRecv(elem(b, b.src, H - d, 0), rowsB, ncclChar, N,
+                            comm_nccl_, comm_));
+      }
+      if (Sn >= 0) {
+        NCCL_CHECK(ncclSend(elem(b, b.src, H + b.g.h - d, 0), rowsB, ncclChar,
+                            Sn, comm_nccl_, comm_));
+        NCCL_CHECK(ncclRecv(elem(b, b.src, H + b.g.h, 0), rowsB, ncclChar, Sn,
+                            comm_nccl_, comm_));
+      }
+      NCCL_CHECK(ncclGroupEnd());
+    }
+  }
+
+  DType dt_;
+  int k_;
+  int S_;
+  int rank_ = 0;
+  std::vector<DeviceBlock> blocks_;
+  hipStream_t compute_ = nullptr;
+  hipStream_t comm_ = nullptr;
+  hipEvent_t evComm_ = nullptr;
+  hipEvent_t evCompute_ = nullptr;
+  void* wgt_ = nullptr;
+  void* diffDev_ = nullptr;
+  ncclComm_t comm_nccl_ = nullptr;
+  bool swapped_ = false;
+};
+
+}  // namespace pconv

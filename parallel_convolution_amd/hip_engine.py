@@ -1,0 +1,166 @@
+"""Python wrapper of the C++/HIP/RCCL engine (csrc/).
+
+The thin PyTorch-ROCm front end of BASELINE.json:5: torch owns the device
+buffers (its caching allocator), the C++ ConvEngine owns streams, the
+iteration loop and the RCCL communicator. This module never touches pixels
+on the hot path — run() is one C++ call for the whole timed loop.
+
+The extension is REQUIRED on a GPU box: if `parallel_convolution_amd._C`
+is missing there we raise instead of silently falling back to eager
+PyTorch (the native path must be the one that runs).
+"""
+
+from __future__ import annotations
+
+import dataclasses
+from typing import List, Optional
+
+import numpy as np
+import torch
+
+from .parallel.decomp import BlockSpec, make_blocks, max_chunk_iters
+from .spec import DType, ProblemSpec
+
+_DT_CODE = {DType.U8: 0, DType.F16: 1, DType.F32: 2}
+
+
+def load_extension():
+    try:
+        from . import _C  # noqa: F401
+
+        return _C
+    except ImportError as e:
+        raise ImportError(
+            "parallel_convolution_amd._C is not built. Run "
+            "`PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace` "
+            f"(import error: {e})"
+        ) from e
+
+
+@dataclasses.dataclass
+class _HipBlock:
+    block: BlockSpec
+    src: torch.Tensor  # (bufh, pitchE) storage dtype
+    dst: torch.Tensor
+
+
+class HipEngine:
+    """One process's view of the run: 1 block (distributed/RCCL) or the
+    whole decomposition (single-process local mode, D2D exchange)."""
+
+    def __init__(
+        self,
+        spec: ProblemSpec,
+        blocks: List[BlockSpec],
+        world_blocks: Optional[List[BlockSpec]] = None,
+        device: str = "cuda",
+        chunk_iters: Optional[int] = None,
+    ):
+        self._C = load_extension()
+        self.spec = spec
+        self.device = torch.device(device)
+        self.blocks: List[_HipBlock] = []
+        filt = [float(v) for v in spec.filter.flat]
+        self.engine = self._C.ConvEngine(
+            _DT_CODE[spec.dtype], spec.k, filt, spec.channels
+        )
+        for bs in blocks:
+            shape = (bs.buf_h, bs.buf_w_elems)
+            src = torch.zeros(shape, dtype=spec.dtype.torch_dtype, device=self.device)
+            dst = torch.zeros(shape, dtype=spec.dtype.torch_dtype, device=self.device)
+            nbr = [
+                -1 if bs.north is None else bs.north,
+                -1 if bs.south is None else bs.south,
+                -1 if bs.west is None else bs.west,
+                -1 if bs.east is None else bs.east,
+            ]
+            idx = self.engine.add_block(
+                src, dst, bs.h, bs.w, bs.halo, spec.channels, nbr
+            )
+            assert idx == len(self.blocks)
+            self.blocks.append(_HipBlock(bs, src, dst))
+        # chunk T must be identical on every rank: derive from the WORLD
+        # decomposition, not just local blocks.
+        wb = world_blocks if world_blocks is not None else blocks
+        t_cap = max_chunk_iters(wb, spec.radius, wb[0].halo)
+        self.chunk_iters = min(chunk_iters or t_cap, t_cap)
+        self._iters_done = 0
+
+    # -- local-mode neighbor remap: BlockSpec neighbors are ranks; in local
+    # mode ranks == local block indices already (make_blocks order), so
+    # nothing to remap.
+
+    def init_distributed(self, rank: int, world: int) -> None:
+        """Create the RCCL communicator; ncclUniqueId broadcast over the
+        (gloo) control group."""
+        import torch.distributed as dist
+
+        from .parallel.bootstrap import broadcast_bytes
+
+        uid = self._C.get_nccl_unique_id() if rank == 0 else None
+        uid = broadcast_bytes(uid, src=0, length=128)
+        assert dist.get_rank() == rank and dist.get_world_size() == world
+        self.engine.init_nccl(uid, rank, world)
+
+    @property
+    def _out_is_dst(self) -> bool:
+        return bool(self.engine.swapped())
+
+    def load_block(self, i: int, img_block: np.ndarray) -> None:
+        hb = self.blocks[i]
+        b, S = hb.block, self.spec.channels
+        t = torch.from_numpy(np.ascontiguousarray(img_block))
+        assert t.shape == (b.h, b.w * S), (t.shape, b)
+        target = hb.dst if self._out_is_dst else hb.src
+        target[b.halo : b.halo + b.h, b.halo * S : (b.halo + b.w) * S].copy_(
+            t.to(self.device)
+        )
+
+    def store_block(self, i: int) -> np.ndarray:
+        hb = self.blocks[i]
+        b, S = hb.block, self.spec.channels
+        t = hb.dst if self._out_is_dst else hb.src
+        out = t[b.halo : b.halo + b.h, b.halo * S : (b.halo + b.w) * S]
+        return out.contiguous().cpu().numpy()
+
+    def run(self, iters: Optional[int] = None) -> None:
+        """Enqueue + complete `iters` iterations (C++ loop; returns after
+        device sync)."""
+        iters = self.spec.iters if iters is None else iters
+        if iters <= 0:
+            return
+        self.engine.run(iters, self.chunk_iters)
+        self._iters_done += iters
+
+    def diff_max(self) -> float:
+        """max |last - previous| over owned regions of local blocks (the
+        convergence test of SURVEY.md §2.1 R10; allreduce is the caller's
+        job in distributed mode)."""
+        return float(self.engine.diff_max())
+
+    def synchronize(self) -> None:
+        self.engine.synchronize()
+
+
+def run_single_gpu(spec: ProblemSpec, image: np.ndarray, device: str = "cuda") -> np.ndarray:
+    """Whole image on one GPU (BASELINE.json:8 config 2 path)."""
+    blocks = make_blocks(spec.height, spec.width, 1, spec.channels)
+    eng = HipEngine(spec, blocks)
+    eng.load_block(0, image)
+    eng.run()
+    return eng.store_block(0)
+
+
+def run_local_blocks_gpu(spec: ProblemSpec, image: np.ndarray, p: int) -> np.ndarray:
+    """P blocks on ONE GPU with D2D exchange — the 1-GPU-box test mode for
+    the multi-GPU protocol."""
+    specs = make_blocks(spec.height, spec.width, p, spec.channels)
+    eng = HipEngine(spec, specs)
+    S = spec.channels
+    for i, bs in enumerate(specs):
+        eng.load_block(i, image[bs.y0 : bs.y0 + bs.h, bs.x0 * S : (bs.x0 + bs.w) * S])
+    eng.run()
+    out = np.zeros_like(image)
+    for i, bs in enumerate(specs):
+        out[bs.y0 : bs.y0 + bs.h, bs.x0 * S : (bs.x0 + bs.w) * S] = eng.store_block(i)
+    return out
