@@ -5,64 +5,78 @@
 // R2-R4 / §2.4 E1-E2; the reference tree was unmounted — SURVEY.md §0 —
 // so the behavioral contract is BASELINE.json:5 and spec.py).
 //
-// Design (guided by /opt/skills/guides/cdna_hip_programming.md):
-//   * 256-thread workgroups (4 waves of 64), tile = 128 output ELEMENTS
-//     (x) × 32 rows; each thread computes 8 consecutive elements × 2 rows.
+// Design (guided by /opt/skills/guides/cdna_hip_programming.md, and by
+// rocprofv3 PMC evidence from the v1 kernel — see profiles/):
+//   * 256-thread workgroups (4 waves of 64); tile = 256 output ELEMENTS
+//     (x) × 8 rows. Threads are laid out 64×4 so ONE WAVE = ONE ROW of
+//     4-element output chunks: every ds_read_b128 has all 64 lanes on the
+//     same LDS row reading contiguous float4s — structurally
+//     bank-conflict-free (the v1 layout put 4 different rows in each read
+//     instruction: SQ_LDS_BANK_CONFLICT was 84 % of LDS-active cycles).
+//   * The LDS tile is a float4 array, indexed in float4 units, so the
+//     compiler can prove 16-B alignment and emit ds_read_b128 /
+//     ds_write_b128 (a float* cast compiled to ds_read2_b32 — 2× LDS
+//     cycles and 2× instructions).
 //   * Input staged through LDS as fp32: global loads are 16-byte
 //     vectorized (Guideline 13), converted once per element (u8 →
-//     v_cvt_f32_ubyte*, f16 → cvt_f32_f16), so each element's 9/25 taps
-//     re-read converted data from LDS instead of re-converting.
-//   * LDS rows padded to 164 dwords (160-dword window + 4) so the
-//     ds_read_b128 lane groups land on distinct banks (Guideline 4; row
-//     stride ≡ 36 mod 64).
+//     v_cvt_f32_ubyte*), so each element's k² taps re-read converted data
+//     instead of re-converting.
 //   * fp32 accumulation in row-major tap order; storage rounding per
 //     spec.py (u8: round-half-even + clamp via __float2int_rn).
 //   * Grayscale and packed-RGB share one template: the channel count S is
 //     the x-stride of a tap in ELEMENT space (interleaved storage,
-//     BASELINE.json:5), so RGB needs no de-interleave pass at all.
-//   * Tiles snap to 16-element x-origins so every staging load is
-//     16B-aligned regardless of the (halo-deepened) region offset.
-//
-// The memory-bound roofline at the headline config is ≈3.1 Tpx-it/s per
-// GPU; the arithmetic here budgets ≈12-14 VALU ops per
-// element so the kernel stays memory-bound.
+//     BASELINE.json:5) — RGB needs no de-interleave pass at all.
+//   * Tiles snap to 16-element x-origins so every staging load stays
+//     16-B-aligned regardless of the (halo-deepened) region offset.
 
 #include "common.hpp"
 
 #include <hip/hip_fp16.h>
 
+#include <algorithm>
+
 namespace pconv {
 
 namespace {
 
-constexpr int TILE_X = 128;  // output elements per tile (x)
-constexpr int TILE_Y = 32;   // output rows per tile
+constexpr int TILE_X = 256;  // output elements per tile (x)
+constexpr int TILE_Y = 8;    // output rows per tile
 constexpr int NTHREADS = 256;
-constexpr int WIN_X = TILE_X + 32;       // staged window elements
-constexpr int LDS_PITCH = WIN_X + 4;     // +4 dwords: bank-conflict pad
+constexpr int VEC = 4;                    // output elements per thread (x)
+constexpr int WIN_X = TILE_X + 32;        // staged window elements
+constexpr int LDS_P4 = WIN_X / 4 + 1;     // LDS row pitch in float4 (+1 pad)
 constexpr int CHUNKS_PER_ROW = WIN_X / 16;
+
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+// LDS-addrspace volatile vector: volatile forbids LLVM from narrowing or
+// splitting the load; address_space(3) keeps it a ds_read_b128 (a generic
+// volatile pointer lowers to flat_load instead).
+typedef __attribute__((address_space(3))) volatile f32x4 lds_v4;
 
 // ---- dtype traits ---------------------------------------------------------
 
 struct U8T {
   using T = unsigned char;
-  static __device__ __forceinline__ void load16(const T* p, float* f) {
-    uint4 v = *reinterpret_cast<const uint4*>(p);
+  using Raw = uint4;  // 16 elements = 16 bytes
+  static constexpr int STORE_W = 4;  // bytes per vector store
+  static __device__ __forceinline__ Raw loadRaw(const T* p) {
+    return *reinterpret_cast<const uint4*>(p);
+  }
+  static __device__ __forceinline__ void cvtWrite(f32x4* ldsp, Raw v) {
     const unsigned* d = reinterpret_cast<const unsigned*>(&v);
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-#pragma unroll
-      for (int b = 0; b < 4; ++b)
-        f[4 * i + b] = (float)((d[i] >> (8 * b)) & 0xffu);
+      f32x4 o = {(float)(d[i] & 0xffu), (float)((d[i] >> 8) & 0xffu),
+                 (float)((d[i] >> 16) & 0xffu), (float)(d[i] >> 24)};
+      ldsp[i] = o;
     }
   }
   static __device__ __forceinline__ unsigned q1(float v) {
     return (unsigned)__float2int_rn(fminf(fmaxf(v, 0.f), 255.f));
   }
-  static __device__ __forceinline__ void store8(T* p, const float* a) {
-    unsigned lo = q1(a[0]) | (q1(a[1]) << 8) | (q1(a[2]) << 16) | (q1(a[3]) << 24);
-    unsigned hi = q1(a[4]) | (q1(a[5]) << 8) | (q1(a[6]) << 16) | (q1(a[7]) << 24);
-    *reinterpret_cast<uint2*>(p) = make_uint2(lo, hi);
+  static __device__ __forceinline__ void store4(T* p, const float* a) {
+    *reinterpret_cast<unsigned*>(p) =
+        q1(a[0]) | (q1(a[1]) << 8) | (q1(a[2]) << 16) | (q1(a[3]) << 24);
   }
   static __device__ __forceinline__ void store1(T* p, float v) { *p = (T)q1(v); }
   static __device__ __forceinline__ float toF(T v) { return (float)v; }
@@ -70,19 +84,29 @@ struct U8T {
 
 struct F16T {
   using T = __half;
-  static __device__ __forceinline__ void load16(const T* p, float* f) {
-    const __half2* h = reinterpret_cast<const __half2*>(p);
+  struct Raw { uint4 a, b; };  // 16 halves = 32 bytes
+  static constexpr int STORE_W = 8;
+  static __device__ __forceinline__ Raw loadRaw(const T* p) {
+    const uint4* q = reinterpret_cast<const uint4*>(p);
+    return {q[0], q[1]};
+  }
+  static __device__ __forceinline__ void cvtWrite(f32x4* ldsp, Raw v) {
+    const __half2* h = reinterpret_cast<const __half2*>(&v);
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      float2 v = __half22float2(h[i]);
-      f[2 * i] = v.x;
-      f[2 * i + 1] = v.y;
+    for (int i = 0; i < 4; ++i) {
+      float2 lo = __half22float2(h[2 * i]);
+      float2 hi = __half22float2(h[2 * i + 1]);
+      f32x4 o = {lo.x, lo.y, hi.x, hi.y};
+      ldsp[i] = o;
     }
   }
-  static __device__ __forceinline__ void store8(T* p, const float* a) {
-    __half2* h = reinterpret_cast<__half2*>(p);
-#pragma unroll
-    for (int i = 0; i < 4; ++i) h[i] = __floats2half2_rn(a[2 * i], a[2 * i + 1]);
+  static __device__ __forceinline__ void store4(T* p, const float* a) {
+    __half2 h0 = __floats2half2_rn(a[0], a[1]);
+    __half2 h1 = __floats2half2_rn(a[2], a[3]);
+    uint2 u;
+    u.x = *reinterpret_cast<unsigned*>(&h0);
+    u.y = *reinterpret_cast<unsigned*>(&h1);
+    *reinterpret_cast<uint2*>(p) = u;
   }
   static __device__ __forceinline__ void store1(T* p, float v) { *p = __float2half_rn(v); }
   static __device__ __forceinline__ float toF(T v) { return __half2float(v); }
@@ -90,21 +114,18 @@ struct F16T {
 
 struct F32T {
   using T = float;
-  static __device__ __forceinline__ void load16(const T* p, float* f) {
-    const float4* q = reinterpret_cast<const float4*>(p);
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      float4 v = q[i];
-      f[4 * i] = v.x;
-      f[4 * i + 1] = v.y;
-      f[4 * i + 2] = v.z;
-      f[4 * i + 3] = v.w;
-    }
+  struct Raw { f32x4 q[4]; };  // 16 floats = 64 bytes
+  static constexpr int STORE_W = 16;
+  static __device__ __forceinline__ Raw loadRaw(const T* p) {
+    const f32x4* q = reinterpret_cast<const f32x4*>(p);
+    return {q[0], q[1], q[2], q[3]};
   }
-  static __device__ __forceinline__ void store8(T* p, const float* a) {
-    float4* q = reinterpret_cast<float4*>(p);
-    q[0] = make_float4(a[0], a[1], a[2], a[3]);
-    q[1] = make_float4(a[4], a[5], a[6], a[7]);
+  static __device__ __forceinline__ void cvtWrite(f32x4* ldsp, Raw v) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) ldsp[i] = v.q[i];
+  }
+  static __device__ __forceinline__ void store4(T* p, const float* a) {
+    *reinterpret_cast<float4*>(p) = make_float4(a[0], a[1], a[2], a[3]);
   }
   static __device__ __forceinline__ void store1(T* p, float v) { *p = v; }
   static __device__ __forceinline__ float toF(T v) { return v; }
@@ -113,102 +134,126 @@ struct F32T {
 // ---- fast LDS-tiled kernel ------------------------------------------------
 
 // R: filter radius; S: element stride of one pixel (1 gray, 3 packed RGB).
+//
+// Pipelined 2-phase structure (cdna_hip_programming.md §5.5 T3 minimum
+// 2-phase + T14 issue-early/write-late): each workgroup owns a CHAIN of
+// y-tiles; while computing tile t from LDS buffer t&1 it has the raw
+// global loads for tile t+1 already in flight (issued before the ds_reads)
+// and converts+writes them to the other buffer after the compute — one
+// barrier per tile. This hides the ~900-cycle HBM/L2 latency that made
+// the one-shot version spend 42 % of wave cycles in SQ_WAIT_ANY.
 template <typename TT, int R, int S>
 __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
     const typename TT::T* __restrict__ src, typename TT::T* __restrict__ dst,
     const float* __restrict__ wgt, int pitchE, int bufh, int ex0, int ex1,
-    int py0, int py1, int abase, int tilesX) {
+    int py0, int py1, int abase, int tilesX, int chainLen) {
   using T = typename TT::T;
+  using Raw = typename TT::Raw;
   constexpr int K = 2 * R + 1;
   constexpr int ROWS_IN = TILE_Y + 2 * R;
-  // aligned window geometry: out element e reads [e-S*R, e+S*R]; with 8
-  // elements per thread the aligned fp32 window is NB 4-dword blocks
-  // starting A0 dwords left of the first output element.
-  constexpr int A0 = ((S * R + 3) / 4) * 4;
-  constexpr int NB = (A0 + 8 + S * R + 3) / 4;
+  constexpr int SR = S * R;
+  constexpr int NL = (SR + 3) / 4;
+  constexpr int NB = 1 + 2 * NL;
+  constexpr int ITEMS = ROWS_IN * CHUNKS_PER_ROW;  // <= 256 for R <= 3
+  static_assert(ITEMS <= NTHREADS, "one staging chunk per thread");
 
-  __shared__ float lds[ROWS_IN][LDS_PITCH];
+  __shared__ f32x4 lds[2][ROWS_IN][LDS_P4];
 
   const int tid = threadIdx.x;
   const int bx = blockIdx.x % tilesX;
-  const int by = blockIdx.x / tilesX;
+  const int bc = blockIdx.x / tilesX;
   const int etx0 = abase + TILE_X * bx;  // tile origin (16-aligned element)
   const int gx0 = etx0 - 16;             // staged window origin
-  const int ty0 = py0 + TILE_Y * by;
-  const int gy0 = ty0 - R;
+  const int tyBase = py0 + TILE_Y * chainLen * bc;
+  const int nt = min(chainLen, (py1 - tyBase + TILE_Y - 1) / TILE_Y);
 
-  // filter → registers (wave-uniform address → scalar loads)
+  // filter -> registers (wave-uniform address -> scalar loads)
   float w[K * K];
 #pragma unroll
   for (int i = 0; i < K * K; ++i) w[i] = wgt[i];
 
-  // stage window (ROWS_IN rows × WIN_X elements) into LDS as fp32
-  for (int i = tid; i < ROWS_IN * CHUNKS_PER_ROW; i += NTHREADS) {
-    const int rr = i / CHUNKS_PER_ROW;
-    const int cc = i % CHUNKS_PER_ROW;
-    const int gy = min(gy0 + rr, bufh - 1);
-    int ga = gx0 + 16 * cc;
-    ga = max(0, min(ga, pitchE - 16));
-    float f[16];
-    TT::load16(src + (size_t)gy * pitchE + ga, f);
-    float4* out = reinterpret_cast<float4*>(&lds[rr][16 * cc]);
-#pragma unroll
-    for (int q = 0; q < 4; ++q)
-      out[q] = make_float4(f[4 * q], f[4 * q + 1], f[4 * q + 2], f[4 * q + 3]);
+  // staging role: one 16-element chunk per thread
+  const int rr = tid / CHUNKS_PER_ROW;
+  const int cc = tid % CHUNKS_PER_ROW;
+  const int ga = max(0, min(gx0 + 16 * cc, pitchE - 16));
+  const bool stager = tid < ITEMS;
+
+  // prologue: stage tile 0
+  if (stager) {
+    const int gy = min(max(tyBase - R + rr, 0), bufh - 1);
+    Raw raw = TT::loadRaw(src + (size_t)gy * pitchE + ga);
+    TT::cvtWrite(&lds[0][rr][4 * cc], raw);
   }
   __syncthreads();
 
-  // each thread: 8 consecutive elements × 2 rows
-  const int tx = tid % 16;
-  const int tyy = tid / 16;
-  const int ex = etx0 + 8 * tx;
-  const int lx = (16 + 8 * tx) - A0;  // LDS col of the window block start
+  // compute lane mapping: one wave per output row-pair
+  const int tx = tid & 63;
+  const int tyy = tid >> 6;
+  const int ex = etx0 + VEC * tx;
+  const int c4 = 4 + tx - NL;  // first float4 col of this thread's window
 
-  float win[2 + 2 * R][NB * 4];
-#pragma unroll
-  for (int rr = 0; rr < 2 + 2 * R; ++rr) {
-    const float4* row =
-        reinterpret_cast<const float4*>(&lds[2 * tyy + rr][lx]);
-#pragma unroll
-    for (int b = 0; b < NB; ++b) {
-      float4 v = row[b];
-      win[rr][4 * b] = v.x;
-      win[rr][4 * b + 1] = v.y;
-      win[rr][4 * b + 2] = v.z;
-      win[rr][4 * b + 3] = v.w;
-    }
-  }
-
-  float acc[2][8];
-#pragma unroll
-  for (int j = 0; j < 2; ++j)
-#pragma unroll
-    for (int p = 0; p < 8; ++p) acc[j][p] = 0.f;
-
-#pragma unroll
-  for (int dy = 0; dy < K; ++dy)
-#pragma unroll
-    for (int dx = 0; dx < K; ++dx) {
-      const float c = w[dy * K + dx];
-#pragma unroll
-      for (int j = 0; j < 2; ++j)
-#pragma unroll
-        for (int p = 0; p < 8; ++p)
-          acc[j][p] = fmaf(c, win[j + dy][A0 - S * R + p + S * dx], acc[j][p]);
+  for (int t = 0; t < nt; ++t) {
+    // issue next tile's global loads BEFORE the ds_reads of this tile
+    Raw pre;
+    const bool pf = stager && (t + 1 < nt);
+    if (pf) {
+      const int gy = min(tyBase + (t + 1) * TILE_Y - R + rr, bufh - 1);
+      pre = TT::loadRaw(src + (size_t)gy * pitchE + ga);
     }
 
+    const int ty0 = tyBase + t * TILE_Y;
+    float win[2 + 2 * R][NB * 4];
 #pragma unroll
-  for (int j = 0; j < 2; ++j) {
-    const int y = ty0 + 2 * tyy + j;
-    if (y >= py1) continue;
-    T* out = dst + (size_t)y * pitchE + ex;
-    if (ex >= ex0 && ex + 8 <= ex1) {
-      TT::store8(out, acc[j]);
-    } else {
+    for (int rw = 0; rw < 2 + 2 * R; ++rw) {
 #pragma unroll
-      for (int p = 0; p < 8; ++p)
-        if (ex + p >= ex0 && ex + p < ex1) TT::store1(out + p, acc[j][p]);
+      for (int b = 0; b < NB; ++b) {
+        // volatile LDS-addrspace vector load: stays one conflict-free
+        // ds_read_b128 (plain loads get unused lanes DCE'd and split into
+        // strided ds_read2_b32 -- 4-way bank conflicts).
+        f32x4 v = *(lds_v4*)&lds[t & 1][2 * tyy + rw][c4 + b];
+        win[rw][4 * b] = v.x;
+        win[rw][4 * b + 1] = v.y;
+        win[rw][4 * b + 2] = v.z;
+        win[rw][4 * b + 3] = v.w;
+      }
     }
+
+    float acc[2][VEC];
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int p = 0; p < VEC; ++p) acc[j][p] = 0.f;
+
+#pragma unroll
+    for (int dy = 0; dy < K; ++dy)
+#pragma unroll
+      for (int dx = 0; dx < K; ++dx) {
+        const float c = w[dy * K + dx];
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+#pragma unroll
+          for (int p = 0; p < VEC; ++p)
+            acc[j][p] =
+                fmaf(c, win[j + dy][4 * NL - SR + p + S * dx], acc[j][p]);
+      }
+
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int y = ty0 + 2 * tyy + j;
+      if (y >= py1) continue;
+      T* out = dst + (size_t)y * pitchE + ex;
+      if (ex >= ex0 && ex + VEC <= ex1) {
+        TT::store4(out, acc[j]);
+      } else {
+#pragma unroll
+        for (int p = 0; p < VEC; ++p)
+          if (ex + p >= ex0 && ex + p < ex1) TT::store1(out + p, acc[j][p]);
+      }
+    }
+
+    // write-late: convert + store the prefetched tile, then one barrier
+    if (pf) TT::cvtWrite(&lds[(t + 1) & 1][rr][4 * cc], pre);
+    __syncthreads();
   }
 }
 
@@ -270,11 +315,16 @@ void launch_tile(const void* src, void* dst, const float* wgt,
   const int abase = (r.ex0 / 16) * 16;
   const int tilesX = (r.ex1 - abase + TILE_X - 1) / TILE_X;
   const int tilesY = (r.py1 - r.py0 + TILE_Y - 1) / TILE_Y;
-  hipLaunchKernelGGL((conv_tile_k<TT, R, S>), dim3(tilesX * tilesY),
+  // chain length: keep >= ~4096 workgroups in flight (256 CUs x 8 blocks
+  // with headroom); longer chains amortize the staging prologue.
+  int chainLen = std::max(1, std::min(16, (tilesX * tilesY) / 4096));
+  const int chains = (tilesY + chainLen - 1) / chainLen;
+  hipLaunchKernelGGL((conv_tile_k<TT, R, S>), dim3(tilesX * chains),
                      dim3(NTHREADS), 0, stream,
                      reinterpret_cast<const typename TT::T*>(src),
                      reinterpret_cast<typename TT::T*>(dst), wgt, g.pitchE,
-                     g.bufh, r.ex0, r.ex1, r.py0, r.py1, abase, tilesX);
+                     g.bufh, r.ex0, r.ex1, r.py0, r.py1, abase, tilesX,
+                     chainLen);
 }
 
 template <typename TT>
@@ -283,6 +333,7 @@ void launch_dtype(int S, int k, const void* src, void* dst, const float* wgt,
   const int R = k / 2;
   if (R == 1 && S == 1) return launch_tile<TT, 1, 1>(src, dst, wgt, g, r, stream);
   if (R == 2 && S == 1) return launch_tile<TT, 2, 1>(src, dst, wgt, g, r, stream);
+  if (R == 3 && S == 1) return launch_tile<TT, 3, 1>(src, dst, wgt, g, r, stream);
   if (R == 1 && S == 3) return launch_tile<TT, 1, 3>(src, dst, wgt, g, r, stream);
   if (R == 2 && S == 3) return launch_tile<TT, 2, 3>(src, dst, wgt, g, r, stream);
   // generic fallback
