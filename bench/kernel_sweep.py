@@ -1,0 +1,53 @@
+"""Direct conv-kernel sweep across image sizes (cache-residency probe).
+
+Sizes chosen so src+dst land in L2 (~32 MB), L3 (~256 MB), or HBM:
+2048²=8 MB, 4096²=33 MB, 8192²=134 MB, 16384²=537 MB (u8, two buffers).
+"""
+
+import sys
+import time
+
+sys.path.insert(0, ".")
+import numpy as np
+import torch
+
+from parallel_convolution_amd.hip_engine import load_extension
+from parallel_convolution_amd.models.filters import GAUSSIAN_BLUR_3
+
+C = load_extension()
+
+
+def sweep(n, dtype=torch.uint8, iters=200, k=3, channels=1):
+    halo = 16
+    pitch = ((n + 2 * halo) * channels + 63) // 64 * 64
+    bufh = n + 2 * halo
+    if dtype == torch.uint8:
+        src = torch.randint(0, 256, (bufh, pitch), dtype=dtype, device="cuda")
+    else:
+        src = (torch.rand((bufh, pitch), device="cuda") * 256).to(dtype)
+    dst = torch.zeros_like(src)
+    wgt = torch.from_numpy(GAUSSIAN_BLUR_3[:k, :k].copy()).cuda().float().flatten()
+    region = [halo, halo + n, halo * channels, (halo + n) * channels]
+    for _ in range(10):
+        C.conv_region(src, dst, wgt, channels, k, halo, region)
+        src, dst = dst, src
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        C.conv_region(src, dst, wgt, channels, k, halo, region)
+        src, dst = dst, src
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    px = float(n) * n * iters
+    print(
+        f"n={n:6d} dtype={str(dtype)[6:]:8s} {px/dt/1e12:.3f} Tpx/s  "
+        f"{dt/iters*1e6:8.1f} us/iter  footprint={2*n*n*src.element_size()/1e6:.0f} MB",
+        flush=True,
+    )
+
+
+if __name__ == "__main__":
+    for n in (2048, 4096, 8192, 16384):
+        sweep(n)
+    sweep(8192, torch.float16)
+    sweep(8192, torch.float32)

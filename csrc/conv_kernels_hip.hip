@@ -57,19 +57,15 @@ typedef __attribute__((address_space(3))) volatile f32x4 lds_v4;
 
 struct U8T {
   using T = unsigned char;
-  using Raw = uint4;  // 16 elements = 16 bytes
+  using Raw = unsigned;  // 4 elements = 4 bytes
   static constexpr int STORE_W = 4;  // bytes per vector store
   static __device__ __forceinline__ Raw loadRaw(const T* p) {
-    return *reinterpret_cast<const uint4*>(p);
+    return *reinterpret_cast<const unsigned*>(p);
   }
-  static __device__ __forceinline__ void cvtWrite(f32x4* ldsp, Raw v) {
-    const unsigned* d = reinterpret_cast<const unsigned*>(&v);
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      f32x4 o = {(float)(d[i] & 0xffu), (float)((d[i] >> 8) & 0xffu),
-                 (float)((d[i] >> 16) & 0xffu), (float)(d[i] >> 24)};
-      ldsp[i] = o;
-    }
+  static __device__ __forceinline__ void cvtWrite(f32x4* ldsp, Raw d) {
+    f32x4 o = {(float)(d & 0xffu), (float)((d >> 8) & 0xffu),
+               (float)((d >> 16) & 0xffu), (float)(d >> 24)};
+    *ldsp = o;
   }
   static __device__ __forceinline__ unsigned q1(float v) {
     return (unsigned)__float2int_rn(fminf(fmaxf(v, 0.f), 255.f));
@@ -84,21 +80,17 @@ struct U8T {
 
 struct F16T {
   using T = __half;
-  struct Raw { uint4 a, b; };  // 16 halves = 32 bytes
+  using Raw = uint2;  // 4 halves = 8 bytes
   static constexpr int STORE_W = 8;
   static __device__ __forceinline__ Raw loadRaw(const T* p) {
-    const uint4* q = reinterpret_cast<const uint4*>(p);
-    return {q[0], q[1]};
+    return *reinterpret_cast<const uint2*>(p);
   }
   static __device__ __forceinline__ void cvtWrite(f32x4* ldsp, Raw v) {
     const __half2* h = reinterpret_cast<const __half2*>(&v);
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      float2 lo = __half22float2(h[2 * i]);
-      float2 hi = __half22float2(h[2 * i + 1]);
-      f32x4 o = {lo.x, lo.y, hi.x, hi.y};
-      ldsp[i] = o;
-    }
+    float2 lo = __half22float2(h[0]);
+    float2 hi = __half22float2(h[1]);
+    f32x4 o = {lo.x, lo.y, hi.x, hi.y};
+    *ldsp = o;
   }
   static __device__ __forceinline__ void store4(T* p, const float* a) {
     __half2 h0 = __floats2half2_rn(a[0], a[1]);
@@ -114,15 +106,13 @@ struct F16T {
 
 struct F32T {
   using T = float;
-  struct Raw { f32x4 q[4]; };  // 16 floats = 64 bytes
+  using Raw = f32x4;  // 4 floats = 16 bytes
   static constexpr int STORE_W = 16;
   static __device__ __forceinline__ Raw loadRaw(const T* p) {
-    const f32x4* q = reinterpret_cast<const f32x4*>(p);
-    return {q[0], q[1], q[2], q[3]};
+    return *reinterpret_cast<const f32x4*>(p);
   }
   static __device__ __forceinline__ void cvtWrite(f32x4* ldsp, Raw v) {
-#pragma unroll
-    for (int i = 0; i < 4; ++i) ldsp[i] = v.q[i];
+    *ldsp = v;
   }
   static __device__ __forceinline__ void store4(T* p, const float* a) {
     *reinterpret_cast<float4*>(p) = make_float4(a[0], a[1], a[2], a[3]);
@@ -154,8 +144,13 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
   constexpr int SR = S * R;
   constexpr int NL = (SR + 3) / 4;
   constexpr int NB = 1 + 2 * NL;
-  constexpr int ITEMS = ROWS_IN * CHUNKS_PER_ROW;  // <= 256 for R <= 3
-  static_assert(ITEMS <= NTHREADS, "one staging chunk per thread");
+  // one 16-B (4-element) LDS quarter per lane per pass: consecutive lanes
+  // write consecutive 16 B, so ds_write_b128 is bank-conflict-free (the
+  // 16-elements-per-lane layout was a 4-way write conflict: lanes wrote
+  // 64 B apart -- 91 M extra LDS cycles per dispatch in rocprofv3).
+  constexpr int QPR = WIN_X / 4;  // staged quarters per row
+  constexpr int ITEMS = ROWS_IN * QPR;
+  constexpr int NITER = (ITEMS + NTHREADS - 1) / NTHREADS;
 
   __shared__ f32x4 lds[2][ROWS_IN][LDS_P4];
 
@@ -172,17 +167,23 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
 #pragma unroll
   for (int i = 0; i < K * K; ++i) w[i] = wgt[i];
 
-  // staging role: one 16-element chunk per thread
-  const int rr = tid / CHUNKS_PER_ROW;
-  const int cc = tid % CHUNKS_PER_ROW;
-  const int ga = max(0, min(gx0 + 16 * cc, pitchE - 16));
-  const bool stager = tid < ITEMS;
+  // staging role: item i covers row i/QPR, quarter i%QPR
+  int srow[NITER], sq[NITER];
+#pragma unroll
+  for (int k = 0; k < NITER; ++k) {
+    const int i = tid + k * NTHREADS;
+    srow[k] = i / QPR;
+    sq[k] = i % QPR;
+  }
 
   // prologue: stage tile 0
-  if (stager) {
-    const int gy = min(max(tyBase - R + rr, 0), bufh - 1);
+#pragma unroll
+  for (int k = 0; k < NITER; ++k) {
+    if (tid + k * NTHREADS >= ITEMS) break;
+    const int gy = min(max(tyBase - R + srow[k], 0), bufh - 1);
+    const int ga = max(0, min(gx0 + 4 * sq[k], pitchE - 4));
     Raw raw = TT::loadRaw(src + (size_t)gy * pitchE + ga);
-    TT::cvtWrite(&lds[0][rr][4 * cc], raw);
+    TT::cvtWrite(&lds[0][srow[k]][sq[k]], raw);
   }
   __syncthreads();
 
@@ -194,11 +195,16 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
 
   for (int t = 0; t < nt; ++t) {
     // issue next tile's global loads BEFORE the ds_reads of this tile
-    Raw pre;
-    const bool pf = stager && (t + 1 < nt);
+    Raw pre[NITER];
+    const bool pf = t + 1 < nt;
     if (pf) {
-      const int gy = min(tyBase + (t + 1) * TILE_Y - R + rr, bufh - 1);
-      pre = TT::loadRaw(src + (size_t)gy * pitchE + ga);
+#pragma unroll
+      for (int k = 0; k < NITER; ++k) {
+        if (tid + k * NTHREADS >= ITEMS) break;
+        const int gy = min(tyBase + (t + 1) * TILE_Y - R + srow[k], bufh - 1);
+        const int ga = max(0, min(gx0 + 4 * sq[k], pitchE - 4));
+        pre[k] = TT::loadRaw(src + (size_t)gy * pitchE + ga);
+      }
     }
 
     const int ty0 = tyBase + t * TILE_Y;
@@ -237,6 +243,18 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
                 fmaf(c, win[j + dy][4 * NL - SR + p + S * dx], acc[j][p]);
       }
 
+    // write-late: convert + LDS-write the prefetched tile BEFORE issuing
+    // this tile's global stores — the vmcnt wait here then only covers the
+    // prefetch loads (which had the whole compute phase to land), not the
+    // stores.
+    if (pf) {
+#pragma unroll
+      for (int k = 0; k < NITER; ++k) {
+        if (tid + k * NTHREADS >= ITEMS) break;
+        TT::cvtWrite(&lds[(t + 1) & 1][srow[k]][sq[k]], pre[k]);
+      }
+    }
+
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
       const int y = ty0 + 2 * tyy + j;
@@ -251,9 +269,13 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
       }
     }
 
-    // write-late: convert + store the prefetched tile, then one barrier
-    if (pf) TT::cvtWrite(&lds[(t + 1) & 1][rr][4 * cc], pre);
-    __syncthreads();
+    // raw barrier with an LDS-only wait: __syncthreads() would also drain
+    // vmcnt(0), i.e. this tile's global STORES (CDNA4 vmcnt counts
+    // stores), serializing every tile behind its store latency
+    // (cdna_hip_programming.md §5 "the step-3 structure's ceiling").
+    // Only the LDS writes must be visible at the barrier.
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
   }
 }
 
