@@ -230,6 +230,16 @@ class CpuEngine:
         t_cap = max_chunk_iters(all_specs, spec.radius, all_specs[0].halo)
         self.chunk_iters = min(chunk_iters or t_cap, t_cap)
 
+    def diff_max(self) -> float:
+        """max |src - dst| over owned regions."""
+        m = 0.0
+        for blk in self.blocks:
+            b = blk.block
+            s = blk.src[:, b.halo : b.halo + b.h, b.halo : b.halo + b.w]
+            d = blk.dst[:, b.halo : b.halo + b.h, b.halo : b.halo + b.w]
+            m = max(m, float((s - d).abs().max()))
+        return m
+
     def run(self, iters: Optional[int] = None) -> RunStats:
         iters = self.spec.iters if iters is None else iters
         r = self.spec.radius
