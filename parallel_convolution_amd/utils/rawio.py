@@ -104,5 +104,14 @@ def synthetic_image(
     shape = (height, width * channels)
     if dtype == DType.U8:
         return rng.integers(0, 256, size=shape, dtype=np.uint8)
-    # float images in [0, 256) — same dynamic range as u8
-    return (rng.random(shape, dtype=np.float32) * 256).astype(dtype.numpy_dtype)
+    # float images in [0, 256) — same dynamic range as u8; generated in
+    # row slabs so the fp32 intermediate never exceeds ~1 GB (the 65536²
+    # fp16 config would otherwise need a 17 GB temporary)
+    out = np.empty(shape, dtype=dtype.numpy_dtype)
+    slab = max(1, (1 << 28) // max(shape[1], 1))
+    for y0 in range(0, height, slab):
+        y1 = min(y0 + slab, height)
+        out[y0:y1] = (rng.random((y1 - y0, shape[1]), dtype=np.float32) * 256).astype(
+            dtype.numpy_dtype
+        )
+    return out
