@@ -39,7 +39,7 @@ pconv::BlockGeom make_geom(int h, int wPx, int haloPx, int S, int pitchE,
 // the given region of a padded (bufh, pitchE) buffer pair.
 void conv_region(torch::Tensor src, torch::Tensor dst, torch::Tensor wgt,
                  int64_t channels, int64_t k, int64_t haloPx,
-                 std::vector<int64_t> region) {
+                 std::vector<int64_t> region, bool use_mfma) {
   TORCH_CHECK(src.is_cuda() && dst.is_cuda() && wgt.is_cuda());
   TORCH_CHECK(src.dim() == 2 && dst.dim() == 2);
   TORCH_CHECK(src.is_contiguous() && dst.is_contiguous());
@@ -60,7 +60,7 @@ void conv_region(torch::Tensor src, torch::Tensor dst, torch::Tensor wgt,
                   (int)region[3]};
   auto stream = at::cuda::getCurrentCUDAStream();
   pconv::launch_conv(dt, (int)channels, (int)k, src.data_ptr(), dst.data_ptr(),
-                     wgt.data_ptr<float>(), g, r, stream.stream());
+                     wgt.data_ptr<float>(), g, r, use_mfma, stream.stream());
   HIP_CHECK(hipGetLastError());
 }
 
@@ -70,9 +70,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native iterative 2D convolution engine (gfx950)";
 
   py::class_<pconv::ConvEngine>(m, "ConvEngine")
-      .def(py::init<int, int, const std::vector<float>&, int>(),
+      .def(py::init<int, int, const std::vector<float>&, int, bool>(),
            py::arg("dtype"), py::arg("k"), py::arg("filter"),
-           py::arg("channels"))
+           py::arg("channels"), py::arg("use_mfma") = false)
       .def("add_block",
            [](pconv::ConvEngine& e, torch::Tensor src, torch::Tensor dst,
               int h, int wPx, int haloPx, int S, std::vector<int> nbr) {
@@ -99,7 +99,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
 
   m.def("conv_region", &conv_region, "single conv kernel over a region",
         py::arg("src"), py::arg("dst"), py::arg("wgt"), py::arg("channels"),
-        py::arg("k"), py::arg("halo"), py::arg("region"));
+        py::arg("k"), py::arg("halo"), py::arg("region"),
+        py::arg("use_mfma") = false);
+
+  m.def("mfma_probe", [](torch::Tensor A, torch::Tensor B) {
+    TORCH_CHECK(A.is_cuda() && B.is_cuda());
+    TORCH_CHECK(A.sizes() == torch::IntArrayRef({16, 32}));
+    TORCH_CHECK(B.sizes() == torch::IntArrayRef({32, 16}));
+    auto D = torch::zeros({16, 16}, A.options());
+    auto stream = at::cuda::getCurrentCUDAStream();
+    pconv::launch_mfma_probe(A.contiguous().data_ptr<float>(),
+                             B.contiguous().data_ptr<float>(),
+                             D.data_ptr<float>(), stream.stream());
+    HIP_CHECK(hipStreamSynchronize(stream.stream()));
+    return D;
+  });
 
   m.def("read_block",
         [](const std::string& path, int64_t imageRowBytes, int64_t xByteOff,

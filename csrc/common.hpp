@@ -59,7 +59,12 @@ struct Region {
 // otherwise.
 void launch_conv(DType dt, int S, int k, const void* src, void* dst,
                  const float* wgt, const BlockGeom& g, const Region& r,
-                 hipStream_t stream);
+                 bool useMfma, hipStream_t stream);
+
+// one v_mfma_f32_16x16x32_f16 with the conv kernel's assumed fragment
+// layout: D[16][16] = A[16][32] . B[32][16] (layout verification probe)
+void launch_mfma_probe(const float* A32, const float* B32, float* D,
+                       hipStream_t stream);
 
 // max |dst - src| over the owned region, accumulated into *resultDev
 // (a single float, caller zeroes it first).

@@ -76,6 +76,27 @@ struct U8T {
   }
   static __device__ __forceinline__ void store1(T* p, float v) { *p = (T)q1(v); }
   static __device__ __forceinline__ float toF(T v) { return (float)v; }
+  static __device__ __forceinline__ void store8(T* p, const float* a) {
+    uint2 u;
+    u.x = q1(a[0]) | (q1(a[1]) << 8) | (q1(a[2]) << 16) | (q1(a[3]) << 24);
+    u.y = q1(a[4]) | (q1(a[5]) << 8) | (q1(a[6]) << 16) | (q1(a[7]) << 24);
+    *reinterpret_cast<uint2*>(p) = u;
+  }
+  // fp16 staging for the MFMA path: 8 bytes -> 8 halves (u8 is exact in fp16)
+  using RawH = uint2;
+  static __device__ __forceinline__ RawH loadRawH(const T* p) {
+    return *reinterpret_cast<const uint2*>(p);
+  }
+  static __device__ __forceinline__ void cvtWriteH(_Float16* p, RawH v) {
+    _Float16 o[8] __attribute__((aligned(16)));
+    const unsigned* d = reinterpret_cast<const unsigned*>(&v);
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int b = 0; b < 4; ++b)
+        o[4 * i + b] = (_Float16)(float)((d[i] >> (8 * b)) & 0xffu);
+    *reinterpret_cast<uint4*>(p) = *reinterpret_cast<const uint4*>(o);
+  }
 };
 
 struct F16T {
@@ -102,6 +123,24 @@ struct F16T {
   }
   static __device__ __forceinline__ void store1(T* p, float v) { *p = __float2half_rn(v); }
   static __device__ __forceinline__ float toF(T v) { return __half2float(v); }
+  static __device__ __forceinline__ void store8(T* p, const float* a) {
+    uint4 u;
+    __half2 h;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      h = __floats2half2_rn(a[2 * i], a[2 * i + 1]);
+      reinterpret_cast<unsigned*>(&u)[i] = *reinterpret_cast<unsigned*>(&h);
+    }
+    *reinterpret_cast<uint4*>(p) = u;
+  }
+  // MFMA staging: pixels are already fp16 — pure 16-B copy
+  using RawH = uint4;
+  static __device__ __forceinline__ RawH loadRawH(const T* p) {
+    return *reinterpret_cast<const uint4*>(p);
+  }
+  static __device__ __forceinline__ void cvtWriteH(_Float16* p, RawH v) {
+    *reinterpret_cast<uint4*>(p) = v;
+  }
 };
 
 struct F32T {
@@ -279,6 +318,202 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
   }
 }
 
+// ---- MFMA path (BASELINE.json north-star: "MFMA fp16/fp32 accumulation") --
+//
+// Formulation: for each filter row dy, the 1-D convolution along x of a
+// 16-row slice is a banded (Toeplitz) matrix product
+//     OUT[16 x 16] += A[16 x 32] . T_dy[32 x 16]
+// computed by one v_mfma_f32_16x16x32_f16: A is the fp16-staged input
+// window (32 elements starting 8 left of the 16 output elements), and
+// T_dy[k][j] = w[dy][dx] where k = j + 8 - S*R + S*dx (zero elsewhere,
+// including the padding the window's unused columns multiply). k x k
+// needs only 2R+1 MFMAs per 256 outputs, so the 9-25 fma/px move from
+// the saturated VALU pipe to the matrix pipe (the two issue
+// concurrently, MI355X_MICROARCH.md "Wave scheduling").
+//
+// Numerics: fp16 inputs with fp32 accumulation. u8 pixels and dyadic
+// filters are exact in fp16, so this path stays byte-exact vs the fp32
+// oracle; it is selected only when every filter coefficient round-trips
+// through fp16 (engine checks; PCONV_DISABLE_MFMA=1 forces the VALU
+// path). f32 storage keeps the VALU path: f32 pixels do not fit fp16,
+// and the exact f32-input MFMA (v_mfma_f32_16x16x4_f32) runs at the f32
+// VALU rate, so a banded-Toeplitz mapping at 3/4 utilization would be a
+// net loss there (see profiles/).
+//
+// LDS: fp16 image, row pitch 192 halves (384 B), 16-B quarters stored at
+// slot q ^ (row & 7) — worst-case 2-way banking on both the staged
+// ds_write_b128 and the A-fragment ds_read_b128 (simulated exhaustively).
+
+typedef _Float16 f16x8v __attribute__((ext_vector_type(8)));
+typedef float f32x4v __attribute__((ext_vector_type(4)));
+
+constexpr int M_TILE_X = 256;  // output elements per workgroup tile
+constexpr int M_TILE_Y = 16;
+constexpr int M_P2 = 320;  // f16 per LDS row (640 B; XOR swizzle stays in-row)
+constexpr int M_QPR = 36;  // staged 16-B quarters per row (288-element window)
+
+template <typename TT, int R, int S>
+__global__ __launch_bounds__(NTHREADS) void conv_mfma_k(
+    const typename TT::T* __restrict__ src, typename TT::T* __restrict__ dst,
+    const float* __restrict__ wgt, int pitchE, int bufh, int ex0, int ex1,
+    int py0, int py1, int abase, int tilesX, int chainLen) {
+  using T = typename TT::T;
+  using RawH = typename TT::RawH;
+  constexpr int K = 2 * R + 1;
+  constexpr int ROWS_IN = M_TILE_Y + 2 * R;
+  constexpr int ITEMS = ROWS_IN * M_QPR;
+  constexpr int NITER = (ITEMS + NTHREADS - 1) / NTHREADS;
+  // C-transpose scratch row pitch (bytes): 64 elements + pad chosen so the
+  // 16-B read-back slots of one lane group land on distinct banks
+  constexpr int CSPB = sizeof(T) == 1 ? 112 : 64 * sizeof(T) + 16;
+
+  struct Smem {
+    _Float16 img[2][ROWS_IN][M_P2];
+    char cs[4][16][CSPB];  // per-wave quantized C transpose
+  };
+  __shared__ Smem sm;
+  auto& lds = sm.img;
+
+  const int tid = threadIdx.x;
+  const int bx = blockIdx.x % tilesX;
+  const int bc = blockIdx.x / tilesX;
+  const int etx0 = abase + M_TILE_X * bx;
+  const int ws = etx0 - 16;  // staged window origin (element)
+  const int tyBase = py0 + M_TILE_Y * chainLen * bc;
+  const int nt = min(chainLen, (py1 - tyBase + M_TILE_Y - 1) / M_TILE_Y);
+
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int jcol = lane & 15;
+  const int kb = lane >> 4;
+
+  // Toeplitz B fragments, one per filter row, built once per workgroup.
+  // B[k][j]: lane holds k = kb*8 + jj for its column jcol.
+  f16x8v bfrag[K];
+#pragma unroll
+  for (int dy = 0; dy < K; ++dy) {
+#pragma unroll
+    for (int jj = 0; jj < 8; ++jj) {
+      const int k = kb * 8 + jj;
+      const int d = k - jcol - 8 + S * R;
+      float v = 0.f;
+      if (d >= 0 && d <= 2 * S * R && d % S == 0) v = wgt[dy * K + d / S];
+      bfrag[dy][jj] = (_Float16)v;
+    }
+  }
+
+  // staging role (fp16 image, 16-B quarter per lane, XOR slot swizzle)
+  int srow[NITER], sq[NITER];
+#pragma unroll
+  for (int k = 0; k < NITER; ++k) {
+    const int i = tid + k * NTHREADS;
+    srow[k] = i / M_QPR;
+    sq[k] = i % M_QPR;
+  }
+
+#pragma unroll
+  for (int k = 0; k < NITER; ++k) {
+    if (tid + k * NTHREADS >= ITEMS) break;
+    const int gy = min(max(tyBase - R + srow[k], 0), bufh - 1);
+    const int ga = max(0, min(ws + 8 * sq[k], pitchE - 8));
+    RawH raw = TT::loadRawH(src + (size_t)gy * pitchE + ga);
+    TT::cvtWriteH(&lds[0][srow[k]][(sq[k] ^ (srow[k] & 7)) * 8], raw);
+  }
+  __syncthreads();
+
+  for (int t = 0; t < nt; ++t) {
+    RawH pre[NITER];
+    const bool pf = t + 1 < nt;
+    if (pf) {
+#pragma unroll
+      for (int k = 0; k < NITER; ++k) {
+        if (tid + k * NTHREADS >= ITEMS) break;
+        const int gy = min(tyBase + (t + 1) * M_TILE_Y - R + srow[k], bufh - 1);
+        const int ga = max(0, min(ws + 8 * sq[k], pitchE - 8));
+        pre[k] = TT::loadRawH(src + (size_t)gy * pitchE + ga);
+      }
+    }
+
+    const int ty0 = tyBase + t * M_TILE_Y;
+    // four 16-element output subtiles per wave (independent MFMA
+    // accumulator chains → latency hiding)
+#pragma unroll
+    for (int sti = 0; sti < 4; ++sti) {
+      const int st = 4 * wave + sti;
+      const int q = 2 * st + 1 + kb;
+      f32x4v acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int dy = 0; dy < K; ++dy) {
+        const int row = jcol + dy;  // A row i = lane&15, plus filter row
+        const f16x8v a = *reinterpret_cast<const f16x8v*>(
+            &lds[t & 1][row][(q ^ (row & 7)) * 8]);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_f16(a, bfrag[dy], acc, 0, 0, 0);
+      }
+      // quantize into the per-wave transpose scratch (accumulator layout:
+      // col = lane&15, row = 4*(lane>>4)+reg — scattered element stores
+      // go to LDS, the global stores below are contiguous 16-B runs)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg)
+        TT::store1(reinterpret_cast<T*>(&sm.cs[wave][4 * kb + reg][0]) +
+                       16 * sti + jcol,
+                   acc[reg]);
+    }
+    {
+      const int r2 = lane >> 2;
+      const int c0 = (lane & 3) * 16;  // 16 consecutive elements per lane
+      const T* srcrow = reinterpret_cast<const T*>(&sm.cs[wave][r2][0]) + c0;
+      const int y = ty0 + r2;
+      const int ex = etx0 + 64 * wave + c0;
+      if (y < py1) {
+        T* out = dst + (size_t)y * pitchE + ex;
+        if (ex >= ex0 && ex + 16 <= ex1) {
+          const uint4* v = reinterpret_cast<const uint4*>(srcrow);
+#pragma unroll
+          for (int i = 0; i < (int)sizeof(T); ++i)
+            reinterpret_cast<uint4*>(out)[i] = v[i];
+        } else {
+#pragma unroll
+          for (int p = 0; p < 16; ++p)
+            if (ex + p >= ex0 && ex + p < ex1) out[p] = srcrow[p];
+        }
+      }
+    }
+
+    if (pf) {
+#pragma unroll
+      for (int k = 0; k < NITER; ++k) {
+        if (tid + k * NTHREADS >= ITEMS) break;
+        TT::cvtWriteH(&lds[(t + 1) & 1][srow[k]][(sq[k] ^ (srow[k] & 7)) * 8],
+                      pre[k]);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+}
+
+// A/B fragment-layout probe: D = A.B for one mfma_f32_16x16x32_f16 with
+// A, B built per the layout assumed above; host test checks D against a
+// CPU matmul (asymmetric inputs catch any transpose).
+__global__ void mfma_probe_k(const float* __restrict__ A32,
+                             const float* __restrict__ B32,
+                             float* __restrict__ D) {
+  const int lane = threadIdx.x & 63;
+  const int jcol = lane & 15;
+  const int kb = lane >> 4;
+  f16x8v a, b;
+#pragma unroll
+  for (int jj = 0; jj < 8; ++jj) {
+    a[jj] = (_Float16)A32[jcol * 32 + kb * 8 + jj];   // A[i][k], i=lane&15
+    b[jj] = (_Float16)B32[(kb * 8 + jj) * 16 + jcol]; // B[k][j], j=lane&15
+  }
+  f32x4v acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, acc, 0, 0, 0);
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg)
+    D[(4 * kb + reg) * 16 + jcol] = acc[reg];  // D[row][col]
+}
+
 // ---- generic fallback (any odd k, any channel stride) ---------------------
 
 template <typename TT>
@@ -349,10 +584,42 @@ void launch_tile(const void* src, void* dst, const float* wgt,
                      chainLen);
 }
 
+template <typename TT, int R, int S>
+void launch_mfma(const void* src, void* dst, const float* wgt,
+                 const BlockGeom& g, const Region& r, hipStream_t stream) {
+  const int abase = (r.ex0 / 16) * 16;
+  const int tilesX = (r.ex1 - abase + M_TILE_X - 1) / M_TILE_X;
+  const int tilesY = (r.py1 - r.py0 + M_TILE_Y - 1) / M_TILE_Y;
+  int chainLen = std::max(1, std::min(16, (tilesX * tilesY) / 4096));
+  const int chains = (tilesY + chainLen - 1) / chainLen;
+  hipLaunchKernelGGL((conv_mfma_k<TT, R, S>), dim3(tilesX * chains),
+                     dim3(NTHREADS), 0, stream,
+                     reinterpret_cast<const typename TT::T*>(src),
+                     reinterpret_cast<typename TT::T*>(dst), wgt, g.pitchE,
+                     g.bufh, r.ex0, r.ex1, r.py0, r.py1, abase, tilesX,
+                     chainLen);
+}
+
+template <typename TT>
+struct HasMfma : std::false_type {};
+template <>
+struct HasMfma<U8T> : std::true_type {};
+template <>
+struct HasMfma<F16T> : std::true_type {};
+
 template <typename TT>
 void launch_dtype(int S, int k, const void* src, void* dst, const float* wgt,
-                  const BlockGeom& g, const Region& r, hipStream_t stream) {
+                  const BlockGeom& g, const Region& r, bool useMfma,
+                  hipStream_t stream) {
   const int R = k / 2;
+  if constexpr (HasMfma<TT>::value) {
+    if (useMfma) {
+      if (R == 1 && S == 1) return launch_mfma<TT, 1, 1>(src, dst, wgt, g, r, stream);
+      if (R == 2 && S == 1) return launch_mfma<TT, 2, 1>(src, dst, wgt, g, r, stream);
+      if (R == 1 && S == 3) return launch_mfma<TT, 1, 3>(src, dst, wgt, g, r, stream);
+      if (R == 2 && S == 3) return launch_mfma<TT, 2, 3>(src, dst, wgt, g, r, stream);
+    }
+  }
   if (R == 1 && S == 1) return launch_tile<TT, 1, 1>(src, dst, wgt, g, r, stream);
   if (R == 2 && S == 1) return launch_tile<TT, 2, 1>(src, dst, wgt, g, r, stream);
   if (R == 3 && S == 1) return launch_tile<TT, 3, 1>(src, dst, wgt, g, r, stream);
@@ -372,16 +639,21 @@ void launch_dtype(int S, int k, const void* src, void* dst, const float* wgt,
 
 void launch_conv(DType dt, int S, int k, const void* src, void* dst,
                  const float* wgt, const BlockGeom& g, const Region& r,
-                 hipStream_t stream) {
+                 bool useMfma, hipStream_t stream) {
   if (r.py1 <= r.py0 || r.ex1 <= r.ex0) return;
   switch (dt) {
     case DType::U8:
-      return launch_dtype<U8T>(S, k, src, dst, wgt, g, r, stream);
+      return launch_dtype<U8T>(S, k, src, dst, wgt, g, r, useMfma, stream);
     case DType::F16:
-      return launch_dtype<F16T>(S, k, src, dst, wgt, g, r, stream);
+      return launch_dtype<F16T>(S, k, src, dst, wgt, g, r, useMfma, stream);
     case DType::F32:
-      return launch_dtype<F32T>(S, k, src, dst, wgt, g, r, stream);
+      return launch_dtype<F32T>(S, k, src, dst, wgt, g, r, false, stream);
   }
+}
+
+void launch_mfma_probe(const float* A32, const float* B32, float* D,
+                       hipStream_t stream) {
+  hipLaunchKernelGGL(mfma_probe_k, dim3(1), dim3(64), 0, stream, A32, B32, D);
 }
 
 void launch_diff_max(DType dt, const void* src, const void* dst,

@@ -59,8 +59,9 @@ struct DeviceBlock {
 
 class ConvEngine {
  public:
-  ConvEngine(int dtype, int k, const std::vector<float>& filter, int channels)
-      : dt_(static_cast<DType>(dtype)), k_(k), S_(channels) {
+  ConvEngine(int dtype, int k, const std::vector<float>& filter, int channels,
+             bool useMfma = false)
+      : dt_(static_cast<DType>(dtype)), k_(k), S_(channels), useMfma_(useMfma) {
     if ((int)filter.size() != k * k) throw std::runtime_error("filter size != k*k");
     HIP_CHECK(hipStreamCreateWithFlags(&compute_, hipStreamNonBlocking));
     HIP_CHECK(hipStreamCreateWithFlags(&comm_, hipStreamNonBlocking));
@@ -135,7 +136,7 @@ class ConvEngine {
         const int ext = d - (t + 1) * r;
         for (auto& b : blocks_) {
           launch_conv(dt_, S_, k_, b.src, b.dst, (const float*)wgt_, b.g,
-                      region_for(b.g, ext), compute_);
+                      region_for(b.g, ext), useMfma_, compute_);
         }
         for (auto& b : blocks_) std::swap(b.src, b.dst);
         swapped_ = !swapped_;
@@ -308,6 +309,7 @@ class ConvEngine {
   void* diffDev_ = nullptr;
   ncclComm_t comm_nccl_ = nullptr;
   bool swapped_ = false;
+  bool useMfma_ = false;
 };
 
 }  // namespace pconv

@@ -24,6 +24,25 @@ from .spec import DType, ProblemSpec
 _DT_CODE = {DType.U8: 0, DType.F16: 1, DType.F32: 2}
 
 
+def mfma_eligible(spec: ProblemSpec) -> bool:
+    """MFMA (fp16-input, fp32-accumulate) path: exact only when every
+    filter coefficient round-trips through fp16 (u8/f16 pixels always do;
+    see csrc/conv_kernels.hip MFMA notes). PCONV_DISABLE_MFMA=1 forces
+    the VALU path for A/B comparison."""
+    import os
+
+    if os.environ.get("PCONV_DISABLE_MFMA"):
+        return False
+    if spec.dtype not in (DType.U8, DType.F16):
+        return False
+    if spec.k not in (3, 5) or spec.channels not in (1, 3):
+        return False
+    import torch as _t
+
+    f = _t.from_numpy(spec.filter.copy())
+    return bool((f.half().float() == f).all())
+
+
 def load_extension():
     try:
         from . import _C  # noqa: F401
@@ -61,8 +80,9 @@ class HipEngine:
         self.device = torch.device(device)
         self.blocks: List[_HipBlock] = []
         filt = [float(v) for v in spec.filter.flat]
+        self.use_mfma = mfma_eligible(spec)
         self.engine = self._C.ConvEngine(
-            _DT_CODE[spec.dtype], spec.k, filt, spec.channels
+            _DT_CODE[spec.dtype], spec.k, filt, spec.channels, self.use_mfma
         )
         for bs in blocks:
             shape = (bs.buf_h, bs.buf_w_elems)

@@ -191,3 +191,65 @@ def test_cli_hip_backend(tmp_path):
     got = np.fromfile(out, dtype=np.uint8).reshape(64, 96)
     spec = ProblemSpec(96, 64, 1, DType.U8, GAUSSIAN_BLUR_3, 8)
     np.testing.assert_array_equal(got, run_reference(spec, img))
+
+
+def test_mfma_probe_layout():
+    """Verify the assumed mfma_f32_16x16x32_f16 fragment layouts with
+    asymmetric A and B (catches any transposed mapping)."""
+    rng = np.random.default_rng(11)
+    # small integers: exact in fp16, exact fp32 accumulation
+    A = torch.tensor(rng.integers(-8, 8, (16, 32)), dtype=torch.float32, device="cuda")
+    B = torch.tensor(rng.integers(-8, 8, (32, 16)), dtype=torch.float32, device="cuda")
+    import parallel_convolution_amd._C as C
+
+    D = C.mfma_probe(A, B)
+    ref = (A.cpu() @ B.cpu())
+    np.testing.assert_array_equal(D.cpu().numpy(), ref.numpy())
+
+
+def test_mfma_engine_selected_and_byte_exact():
+    """blur3 is fp16-exact, so the engine must take the MFMA path and stay
+    byte-exact vs the fp32 oracle; the VALU path must agree bitwise."""
+    import os
+
+    from parallel_convolution_amd.hip_engine import HipEngine, mfma_eligible
+    from parallel_convolution_amd.parallel.decomp import make_blocks
+
+    spec = ProblemSpec(512, 384, 1, DType.U8, GAUSSIAN_BLUR_3, 10)
+    assert mfma_eligible(spec)
+    img = synthetic_image(384, 512, 1, DType.U8)
+    out_mfma = _engine_mod().run_single_gpu(spec, img)
+    ref = run_reference(spec, img)
+    np.testing.assert_array_equal(out_mfma, ref)
+    os.environ["PCONV_DISABLE_MFMA"] = "1"
+    try:
+        out_valu = _engine_mod().run_single_gpu(spec, img)
+    finally:
+        del os.environ["PCONV_DISABLE_MFMA"]
+    np.testing.assert_array_equal(out_valu, out_mfma)
+
+
+@pytest.mark.parametrize("W,H,C,dt,filt,iters", [
+    (256, 256, 1, DType.U8, GAUSSIAN_BLUR_5, 6),
+    (256, 200, 1, DType.F16, GAUSSIAN_BLUR_3, 10),
+    (200, 160, 3, DType.U8, GAUSSIAN_BLUR_3, 5),
+    (200, 160, 3, DType.U8, SHARPEN_5, 5),
+    (131, 97, 1, DType.U8, GAUSSIAN_BLUR_3, 7),
+])
+def test_mfma_variants_match_oracle(W, H, C, dt, filt, iters):
+    from parallel_convolution_amd.hip_engine import mfma_eligible
+
+    spec = ProblemSpec(W, H, C, dt, filt, iters)
+    assert mfma_eligible(spec), "these filters are fp16-exact"
+    img = synthetic_image(H, W, C, dt)
+    out = _engine_mod().run_single_gpu(spec, img)
+    ref = run_reference(spec, img)
+    assert_matches(spec, out, ref)
+
+
+def test_mfma_multiblock_protocol():
+    spec = ProblemSpec(96, 80, 1, DType.U8, GAUSSIAN_BLUR_3, 20)
+    img = synthetic_image(80, 96, 1, DType.U8)
+    out = _engine_mod().run_local_blocks_gpu(spec, img, 8)
+    ref = run_reference(spec, img)
+    np.testing.assert_array_equal(out, ref)
