@@ -92,6 +92,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            py::arg("chunk_iters"))
       .def("diff_max", &pconv::ConvEngine::diff_max)
       .def("swapped", &pconv::ConvEngine::swapped)
+      .def("set_phase_timing", &pconv::ConvEngine::set_phase_timing)
+      .def("phase_times", &pconv::ConvEngine::phase_times)
       .def("synchronize", &pconv::ConvEngine::synchronize);
 
   m.def("get_nccl_unique_id",

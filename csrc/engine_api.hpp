@@ -67,6 +67,7 @@ class ConvEngine {
     HIP_CHECK(hipStreamCreateWithFlags(&comm_, hipStreamNonBlocking));
     HIP_CHECK(hipEventCreateWithFlags(&evComm_, hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&evCompute_, hipEventDisableTiming));
+    for (int i = 0; i < 4; ++i) HIP_CHECK(hipEventCreate(&evPhase_[i]));
     HIP_CHECK(hipMalloc(&wgt_, filter.size() * sizeof(float)));
     HIP_CHECK(hipMemcpy(wgt_, filter.data(), filter.size() * sizeof(float),
                         hipMemcpyHostToDevice));
@@ -85,6 +86,7 @@ class ConvEngine {
     if (comm_nccl_) (void)ncclCommDestroy(comm_nccl_);
     (void)hipEventDestroy(evComm_);
     (void)hipEventDestroy(evCompute_);
+    for (int i = 0; i < 4; ++i) (void)hipEventDestroy(evPhase_[i]);
     (void)hipStreamDestroy(compute_);
     (void)hipStreamDestroy(comm_);
   }
@@ -124,14 +126,21 @@ class ConvEngine {
   }
 
   // Run `iters` iterations in chunks of <= chunkT (exchange depth r*T per
-  // chunk). Fully asynchronous; blocks until the device finishes.
+  // chunk). Fully asynchronous; blocks until the device finishes. With
+  // phase timing enabled (per-phase tracing), hipEvent
+  // pairs around each chunk's exchange and compute accumulate into
+  // exchangeMs_/computeMs_ (adds one device sync per chunk — keep it off
+  // in benchmark runs).
   void run(int iters, int chunkT) {
     const int r = k_ / 2;
     int left = iters;
     while (left > 0) {
       const int T = std::min(chunkT, left);
       const int d = r * T;
+      if (phaseTiming_) HIP_CHECK(hipEventRecord(evPhase_[0], comm_));
       if (r > 0 && d > 0) exchange(d);
+      if (phaseTiming_) HIP_CHECK(hipEventRecord(evPhase_[1], comm_));
+      if (phaseTiming_) HIP_CHECK(hipEventRecord(evPhase_[2], compute_));
       for (int t = 0; t < T; ++t) {
         const int ext = d - (t + 1) * r;
         for (auto& b : blocks_) {
@@ -142,10 +151,26 @@ class ConvEngine {
         swapped_ = !swapped_;
       }
       left -= T;
+      if (phaseTiming_) {
+        HIP_CHECK(hipEventRecord(evPhase_[3], compute_));
+        HIP_CHECK(hipEventSynchronize(evPhase_[3]));
+        float ms = 0.f;
+        HIP_CHECK(hipEventElapsedTime(&ms, evPhase_[0], evPhase_[1]));
+        exchangeMs_ += ms;
+        HIP_CHECK(hipEventElapsedTime(&ms, evPhase_[2], evPhase_[3]));
+        computeMs_ += ms;
+        chunks_ += 1;
+      }
     }
     HIP_CHECK(hipStreamSynchronize(compute_));
     HIP_CHECK(hipStreamSynchronize(comm_));
     HIP_CHECK(hipGetLastError());
+  }
+
+  void set_phase_timing(bool on) { phaseTiming_ = on; }
+  // (exchange_ms, compute_ms, chunks) accumulated since construction
+  std::vector<double> phase_times() const {
+    return {exchangeMs_, computeMs_, (double)chunks_};
   }
 
   // max |src - dst| over owned regions of all local blocks (convergence
@@ -310,6 +335,11 @@ class ConvEngine {
   ncclComm_t comm_nccl_ = nullptr;
   bool swapped_ = false;
   bool useMfma_ = false;
+  bool phaseTiming_ = false;
+  hipEvent_t evPhase_[4] = {};
+  double exchangeMs_ = 0.0;
+  double computeMs_ = 0.0;
+  long chunks_ = 0;
 };
 
 }  // namespace pconv

@@ -59,6 +59,8 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--watchdog", type=int, default=0, metavar="SEC",
                    help="abort if an engine segment exceeds SEC seconds")
     p.add_argument("--metrics-out", default=None, help="append JSON run record here")
+    p.add_argument("--phase-timing", action="store_true",
+                   help="per-phase (exchange/compute) hipEvent timing in the record")
     p.add_argument("--no-output", action="store_true", help="skip writing the output image")
     return p
 
@@ -139,6 +141,8 @@ def run_rank(args) -> int:
         eng = HipEngine(spec, [my], world_blocks=blocks, chunk_iters=args.chunk)
         if world > 1:
             eng.init_distributed(rank, world)
+        if args.phase_timing:
+            eng.set_phase_timing(True)
         eng.load_block(0, img_block)
 
         def run_seg(n):
@@ -234,6 +238,11 @@ def run_rank(args) -> int:
                 "converged": converged,
                 "grid": "x".join(map(str, my.grid)),
                 "output": None if args.no_output else out_path,
+                **(
+                    {"phases": eng.phase_times()}
+                    if args.phase_timing and backend == "hip"
+                    else {}
+                ),
             },
         )
         metrics.emit(rec, args.metrics_out)

@@ -158,6 +158,15 @@ class HipEngine:
         job in distributed mode)."""
         return float(self.engine.diff_max())
 
+    def set_phase_timing(self, on: bool) -> None:
+        """hipEvent pairs around each chunk's exchange/compute (SURVEY.md
+        §5.1); adds one device sync per chunk — keep off in timed bench."""
+        self.engine.set_phase_timing(on)
+
+    def phase_times(self) -> dict:
+        ex, comp, chunks = self.engine.phase_times()
+        return {"exchange_ms": ex, "compute_ms": comp, "chunks": int(chunks)}
+
     def synchronize(self) -> None:
         self.engine.synchronize()
 
