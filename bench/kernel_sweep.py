@@ -17,7 +17,7 @@ from parallel_convolution_amd.models.filters import GAUSSIAN_BLUR_3
 C = load_extension()
 
 
-def sweep(n, dtype=torch.uint8, iters=200, k=3, channels=1):
+def sweep(n, dtype=torch.uint8, iters=200, k=3, channels=1, use_mfma=False):
     halo = 16
     pitch = ((n + 2 * halo) * channels + 63) // 64 * 64
     bufh = n + 2 * halo
@@ -29,18 +29,18 @@ def sweep(n, dtype=torch.uint8, iters=200, k=3, channels=1):
     wgt = torch.from_numpy(GAUSSIAN_BLUR_3[:k, :k].copy()).cuda().float().flatten()
     region = [halo, halo + n, halo * channels, (halo + n) * channels]
     for _ in range(10):
-        C.conv_region(src, dst, wgt, channels, k, halo, region)
+        C.conv_region(src, dst, wgt, channels, k, halo, region, use_mfma)
         src, dst = dst, src
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
-        C.conv_region(src, dst, wgt, channels, k, halo, region)
+        C.conv_region(src, dst, wgt, channels, k, halo, region, use_mfma)
         src, dst = dst, src
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     px = float(n) * n * iters
     print(
-        f"n={n:6d} dtype={str(dtype)[6:]:8s} {px/dt/1e12:.3f} Tpx/s  "
+        f"n={n:6d} dtype={str(dtype)[6:]:8s} mfma={int(use_mfma)} {px/dt/1e12:.3f} Tpx/s  "
         f"{dt/iters*1e6:8.1f} us/iter  footprint={2*n*n*src.element_size()/1e6:.0f} MB",
         flush=True,
     )
@@ -49,5 +49,7 @@ def sweep(n, dtype=torch.uint8, iters=200, k=3, channels=1):
 if __name__ == "__main__":
     for n in (2048, 4096, 8192, 16384):
         sweep(n)
+        sweep(n, use_mfma=True)
     sweep(8192, torch.float16)
+    sweep(8192, torch.float16, use_mfma=True)
     sweep(8192, torch.float32)

@@ -88,13 +88,15 @@ struct U8T {
     return *reinterpret_cast<const uint2*>(p);
   }
   static __device__ __forceinline__ void cvtWriteH(_Float16* p, RawH v) {
+    // u8 -> f16 directly via v_cvt_f16_u16 (exact for 0..255): one cvt
+    // per element instead of the 2-op u8->f32->f16 chain
     _Float16 o[8] __attribute__((aligned(16)));
     const unsigned* d = reinterpret_cast<const unsigned*>(&v);
 #pragma unroll
     for (int i = 0; i < 2; ++i)
 #pragma unroll
       for (int b = 0; b < 4; ++b)
-        o[4 * i + b] = (_Float16)(float)((d[i] >> (8 * b)) & 0xffu);
+        o[4 * i + b] = (_Float16)(unsigned short)((d[i] >> (8 * b)) & 0xffu);
     *reinterpret_cast<uint4*>(p) = *reinterpret_cast<const uint4*>(o);
   }
 };
@@ -218,7 +220,8 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
   // prologue: stage tile 0
 #pragma unroll
   for (int k = 0; k < NITER; ++k) {
-    if (tid + k * NTHREADS >= ITEMS) break;
+    if (k == NITER - 1 && ITEMS % NTHREADS != 0 && tid >= ITEMS - k * NTHREADS)
+      break;
     const int gy = min(max(tyBase - R + srow[k], 0), bufh - 1);
     const int ga = max(0, min(gx0 + 4 * sq[k], pitchE - 4));
     Raw raw = TT::loadRaw(src + (size_t)gy * pitchE + ga);
@@ -239,7 +242,8 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
     if (pf) {
 #pragma unroll
       for (int k = 0; k < NITER; ++k) {
-        if (tid + k * NTHREADS >= ITEMS) break;
+        if (k == NITER - 1 && ITEMS % NTHREADS != 0 && tid >= ITEMS - k * NTHREADS)
+      break;
         const int gy = min(tyBase + (t + 1) * TILE_Y - R + srow[k], bufh - 1);
         const int ga = max(0, min(gx0 + 4 * sq[k], pitchE - 4));
         pre[k] = TT::loadRaw(src + (size_t)gy * pitchE + ga);
@@ -289,7 +293,8 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
     if (pf) {
 #pragma unroll
       for (int k = 0; k < NITER; ++k) {
-        if (tid + k * NTHREADS >= ITEMS) break;
+        if (k == NITER - 1 && ITEMS % NTHREADS != 0 && tid >= ITEMS - k * NTHREADS)
+      break;
         TT::cvtWrite(&lds[(t + 1) & 1][srow[k]][sq[k]], pre[k]);
       }
     }
@@ -413,7 +418,8 @@ __global__ __launch_bounds__(NTHREADS) void conv_mfma_k(
 
 #pragma unroll
   for (int k = 0; k < NITER; ++k) {
-    if (tid + k * NTHREADS >= ITEMS) break;
+    if (k == NITER - 1 && ITEMS % NTHREADS != 0 && tid >= ITEMS - k * NTHREADS)
+      break;
     const int gy = min(max(tyBase - R + srow[k], 0), bufh - 1);
     const int ga = max(0, min(ws + 8 * sq[k], pitchE - 8));
     RawH raw = TT::loadRawH(src + (size_t)gy * pitchE + ga);
@@ -427,7 +433,8 @@ __global__ __launch_bounds__(NTHREADS) void conv_mfma_k(
     if (pf) {
 #pragma unroll
       for (int k = 0; k < NITER; ++k) {
-        if (tid + k * NTHREADS >= ITEMS) break;
+        if (k == NITER - 1 && ITEMS % NTHREADS != 0 && tid >= ITEMS - k * NTHREADS)
+      break;
         const int gy = min(tyBase + (t + 1) * M_TILE_Y - R + srow[k], bufh - 1);
         const int ga = max(0, min(ws + 8 * sq[k], pitchE - 8));
         pre[k] = TT::loadRawH(src + (size_t)gy * pitchE + ga);
@@ -482,7 +489,8 @@ __global__ __launch_bounds__(NTHREADS) void conv_mfma_k(
     if (pf) {
 #pragma unroll
       for (int k = 0; k < NITER; ++k) {
-        if (tid + k * NTHREADS >= ITEMS) break;
+        if (k == NITER - 1 && ITEMS % NTHREADS != 0 && tid >= ITEMS - k * NTHREADS)
+      break;
         TT::cvtWriteH(&lds[(t + 1) & 1][srow[k]][(sq[k] ^ (srow[k] & 7)) * 8],
                       pre[k]);
       }

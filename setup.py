@@ -16,6 +16,12 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 
 ROCM_LIB = os.path.join(os.environ.get("ROCM_PATH", "/opt/rocm"), "lib")
 
+# Debug build: PCONV_ASAN=1 adds host AddressSanitizer
+# (run python under LD_PRELOAD=$(gcc -print-file-name=libasan.so)) and -g.
+_asan = bool(os.environ.get("PCONV_ASAN"))
+_dbg_cxx = ["-fsanitize=address", "-g", "-fno-omit-frame-pointer"] if _asan else []
+_dbg_ld = ["-fsanitize=address"] if _asan else []
+
 ext = CUDAExtension(
     name="parallel_convolution_amd._C",
     sources=[
@@ -24,12 +30,12 @@ ext = CUDAExtension(
         "csrc/io.cpp",
     ],
     extra_compile_args={
-        "cxx": ["-O3", "-std=c++17"],
+        "cxx": ["-O3", "-std=c++17"] + _dbg_cxx,
         "nvcc": ["-O3", "-std=c++17"],
     },
     libraries=["rccl"],
     library_dirs=[ROCM_LIB],
-    extra_link_args=[f"-Wl,-rpath,{ROCM_LIB}"],
+    extra_link_args=[f"-Wl,-rpath,{ROCM_LIB}"] + _dbg_ld,
 )
 
 setup(
