@@ -25,10 +25,16 @@ _DT_CODE = {DType.U8: 0, DType.F16: 1, DType.F32: 2}
 
 
 def mfma_eligible(spec: ProblemSpec) -> bool:
-    """MFMA (fp16-input, fp32-accumulate) path: exact only when every
-    filter coefficient round-trips through fp16 (u8/f16 pixels always do;
-    see csrc/conv_kernels.hip MFMA notes). PCONV_DISABLE_MFMA=1 forces
-    the VALU path for A/B comparison."""
+    """MFMA (fp16-input, fp32-accumulate) path selection.
+
+    Numerically valid only when every filter coefficient round-trips
+    through fp16 (u8/f16 pixels always do; see csrc/conv_kernels.hip MFMA
+    notes). Measured A/B (profiles/r01_single_gpu.md): parity with the
+    VALU kernel at k=3 (9 taps — staging-bound either way), +72-83 % at
+    k=5 (25 taps — the VALU path is fma-issue-bound there), so the
+    default picks MFMA for k=5 only. PCONV_DISABLE_MFMA=1 /
+    PCONV_FORCE_MFMA=1 override for A/B.
+    """
     import os
 
     if os.environ.get("PCONV_DISABLE_MFMA"):
@@ -40,7 +46,11 @@ def mfma_eligible(spec: ProblemSpec) -> bool:
     import torch as _t
 
     f = _t.from_numpy(spec.filter.copy())
-    return bool((f.half().float() == f).all())
+    if not bool((f.half().float() == f).all()):
+        return False
+    if os.environ.get("PCONV_FORCE_MFMA"):
+        return True
+    return spec.k == 5
 
 
 def load_extension():
