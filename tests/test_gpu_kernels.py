@@ -208,18 +208,23 @@ def test_mfma_probe_layout():
 
 
 def test_mfma_engine_selected_and_byte_exact():
-    """blur3 is fp16-exact, so the engine must take the MFMA path and stay
-    byte-exact vs the fp32 oracle; the VALU path must agree bitwise."""
+    """MFMA and VALU paths must agree bitwise with the fp32 oracle for a
+    fp16-exact filter (blur3); k=5 filters select MFMA by default."""
     import os
 
-    from parallel_convolution_amd.hip_engine import HipEngine, mfma_eligible
-    from parallel_convolution_amd.parallel.decomp import make_blocks
+    from parallel_convolution_amd.hip_engine import mfma_eligible
 
+    assert mfma_eligible(ProblemSpec(64, 64, 1, DType.U8, GAUSSIAN_BLUR_5, 1))
+    assert mfma_eligible(ProblemSpec(64, 64, 3, DType.U8, SHARPEN_5, 1))
     spec = ProblemSpec(512, 384, 1, DType.U8, GAUSSIAN_BLUR_3, 10)
-    assert mfma_eligible(spec)
     img = synthetic_image(384, 512, 1, DType.U8)
-    out_mfma = _engine_mod().run_single_gpu(spec, img)
     ref = run_reference(spec, img)
+    os.environ["PCONV_FORCE_MFMA"] = "1"
+    try:
+        assert mfma_eligible(spec)
+        out_mfma = _engine_mod().run_single_gpu(spec, img)
+    finally:
+        del os.environ["PCONV_FORCE_MFMA"]
     np.testing.assert_array_equal(out_mfma, ref)
     os.environ["PCONV_DISABLE_MFMA"] = "1"
     try:
@@ -237,19 +242,29 @@ def test_mfma_engine_selected_and_byte_exact():
     (131, 97, 1, DType.U8, GAUSSIAN_BLUR_3, 7),
 ])
 def test_mfma_variants_match_oracle(W, H, C, dt, filt, iters):
-    from parallel_convolution_amd.hip_engine import mfma_eligible
+    """Every (dtype, k, channels) MFMA variant vs the oracle (forced on)."""
+    import os
 
     spec = ProblemSpec(W, H, C, dt, filt, iters)
-    assert mfma_eligible(spec), "these filters are fp16-exact"
     img = synthetic_image(H, W, C, dt)
-    out = _engine_mod().run_single_gpu(spec, img)
+    os.environ["PCONV_FORCE_MFMA"] = "1"
+    try:
+        out = _engine_mod().run_single_gpu(spec, img)
+    finally:
+        del os.environ["PCONV_FORCE_MFMA"]
     ref = run_reference(spec, img)
     assert_matches(spec, out, ref)
 
 
 def test_mfma_multiblock_protocol():
+    import os
+
     spec = ProblemSpec(96, 80, 1, DType.U8, GAUSSIAN_BLUR_3, 20)
     img = synthetic_image(80, 96, 1, DType.U8)
-    out = _engine_mod().run_local_blocks_gpu(spec, img, 8)
+    os.environ["PCONV_FORCE_MFMA"] = "1"
+    try:
+        out = _engine_mod().run_local_blocks_gpu(spec, img, 8)
+    finally:
+        del os.environ["PCONV_FORCE_MFMA"]
     ref = run_reference(spec, img)
     np.testing.assert_array_equal(out, ref)
