@@ -138,10 +138,51 @@ class ConvEngine {
       const int T = std::min(chunkT, left);
       const int d = r * T;
       if (phaseTiming_) HIP_CHECK(hipEventRecord(evPhase_[0], comm_));
-      if (r > 0 && d > 0) exchange(d);
-      if (phaseTiming_) HIP_CHECK(hipEventRecord(evPhase_[1], comm_));
+      int t0 = 0;
+      if (r > 0 && d > 0 && any_neighbor()) {
+        // Overlap (BASELINE.json north-star: halo exchange "overlapped
+        // with interior-tile compute on a second HIP stream"): iteration
+        // 0's INTERIOR — the owned block shrunk by r, whose input
+        // windows never touch the halo — launches on the compute stream
+        // concurrently with the exchange on the comm stream; the four
+        // border strips of region(0) launch after the exchange event.
+        HIP_CHECK(hipEventRecord(evCompute_, compute_));
+        HIP_CHECK(hipStreamWaitEvent(comm_, evCompute_, 0));
+        for (auto& b : blocks_) {
+          launch_conv(dt_, S_, k_, b.src, b.dst, (const float*)wgt_, b.g,
+                      interior_region(b.g), useMfma_, compute_);
+        }
+        if (comm_nccl_) {
+          exchange_nccl(d);
+        } else {
+          exchange_local(d);
+        }
+        if (phaseTiming_) HIP_CHECK(hipEventRecord(evPhase_[1], comm_));
+        HIP_CHECK(hipEventRecord(evComm_, comm_));
+        HIP_CHECK(hipStreamWaitEvent(compute_, evComm_, 0));
+        const int ext0 = d - r;
+        for (auto& b : blocks_) {
+          Region full = region_for(b.g, ext0);
+          Region in = interior_region(b.g);
+          // N / S full-width strips, W / E strips between them
+          Region strips[4] = {
+              {full.py0, in.py0, full.ex0, full.ex1},
+              {in.py1, full.py1, full.ex0, full.ex1},
+              {in.py0, in.py1, full.ex0, in.ex0},
+              {in.py0, in.py1, in.ex1, full.ex1},
+          };
+          for (const Region& s : strips)
+            launch_conv(dt_, S_, k_, b.src, b.dst, (const float*)wgt_, b.g, s,
+                        useMfma_, compute_);
+        }
+        for (auto& b : blocks_) std::swap(b.src, b.dst);
+        swapped_ = !swapped_;
+        t0 = 1;
+      }
+      if (phaseTiming_ && t0 == 0)
+        HIP_CHECK(hipEventRecord(evPhase_[1], comm_));  // no exchange
       if (phaseTiming_) HIP_CHECK(hipEventRecord(evPhase_[2], compute_));
-      for (int t = 0; t < T; ++t) {
+      for (int t = t0; t < T; ++t) {
         const int ext = d - (t + 1) * r;
         for (auto& b : blocks_) {
           launch_conv(dt_, S_, k_, b.src, b.dst, (const float*)wgt_, b.g,
@@ -196,6 +237,26 @@ class ConvEngine {
   }
 
  private:
+  bool any_neighbor() const {
+    for (const auto& b : blocks_)
+      for (int i = 0; i < 4; ++i)
+        if (b.g.nbr[i] >= 0) return true;
+    return false;
+  }
+
+  // iteration-0 interior: the owned block shrunk by r on every side
+  // (empty if the block is thinner than 2r+1 — the border strips then
+  // cover everything)
+  Region interior_region(const BlockGeom& g) const {
+    const int r = k_ / 2;
+    Region rg;
+    rg.py0 = g.haloPx + r;
+    rg.py1 = std::max(g.haloPx + g.h - r, rg.py0);
+    rg.ex0 = (g.haloPx + r) * g.S;
+    rg.ex1 = std::max((g.haloPx + g.wPx - r) * g.S, rg.ex0);
+    return rg;
+  }
+
   Region region_for(const BlockGeom& g, int ext) const {
     Region r;
     r.py0 = g.haloPx - (g.nbr[NB_N] >= 0 ? ext : 0);
