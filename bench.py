@@ -52,6 +52,8 @@ def main() -> int:
     ap.add_argument("--preset", default="headline", choices=sorted(PRESETS))
     ap.add_argument("--chunk", type=int, default=None, help="halo-deepening chunk T")
     ap.add_argument("--json-out", default=None)
+    ap.add_argument("--backend", choices=["hip", "cpu"], default="hip",
+                    help="cpu = gloo protocol path (distributed-flow test only)")
     args = ap.parse_args()
 
     world = bootstrap.env_world()
@@ -74,9 +76,6 @@ def main() -> int:
 
     if world > 1:
         bootstrap.init_control_group("gloo")
-    torch.cuda.set_device(local_rank)
-
-    from parallel_convolution_amd.hip_engine import HipEngine
 
     blocks = make_blocks(spec.height, spec.width, n_gpus, spec.channels)
     if world > 1:
@@ -86,25 +85,60 @@ def main() -> int:
         assert n_gpus == 1, "multi-GPU runs must be launched via torchrun"
         my_blocks = blocks
 
-    eng = HipEngine(spec, my_blocks, world_blocks=blocks, chunk_iters=args.chunk)
-    if world > 1:
-        eng.init_distributed(rank, world)
+    if args.backend == "hip":
+        torch.cuda.set_device(local_rank)
+        from parallel_convolution_amd.hip_engine import HipEngine
+
+        eng = HipEngine(spec, my_blocks, world_blocks=blocks, chunk_iters=args.chunk)
+        if world > 1:
+            eng.init_distributed(rank, world)
+    else:
+        from parallel_convolution_amd.cpu_engine import (
+            CpuBlock,
+            CpuEngine,
+            GlooExchanger,
+            LocalExchanger,
+        )
+
+        cblocks = [CpuBlock(spec, bs) for bs in my_blocks]
+        if world > 1:
+            ex = GlooExchanger(cblocks[0])
+        elif len(cblocks) > 1:
+            ex = LocalExchanger(cblocks)
+        else:
+            ex = None
+        cpu_eng = CpuEngine(spec, cblocks, exchanger=ex, chunk_iters=args.chunk)
+
+        class _CpuShim:
+            chunk_iters = cpu_eng.chunk_iters
+
+            def load_block(self, i, img):
+                cblocks[i].load(img)
+
+            def run(self, n):
+                cpu_eng.run(n)
+
+        eng = _CpuShim()
 
     # synthetic random-pixel block data (BASELINE north-star), per-rank seed
     for i, bs in enumerate(my_blocks):
         eng.load_block(i, synthetic_image(bs.h, bs.w, spec.channels, spec.dtype, seed=1234 + bs.rank))
 
+    def dev_sync():
+        if args.backend == "hip":
+            torch.cuda.synchronize()
+
     # warmup (untimed)
     if args.warmup > 0:
         eng.run(args.warmup)
-    torch.cuda.synchronize()
+    dev_sync()
     if world > 1:
         dist.barrier()
-    torch.cuda.synchronize()
+    dev_sync()
 
     t0 = time.perf_counter()
     eng.run(args.steps)
-    torch.cuda.synchronize()
+    dev_sync()
     t1 = time.perf_counter()
     elapsed = t1 - t0
 
@@ -131,6 +165,7 @@ def main() -> int:
             "vs_baseline": None,
             "dtype": spec.dtype.value,
             "data": "synthetic random-pixel image, seeded per rank",
+            "backend": args.backend,
             "config": {
                 "model": f"iterconv-{spec.k}x{spec.k}-"
                 f"{'gray' if spec.channels == 1 else 'rgb'}-{spec.dtype.value}",

@@ -116,3 +116,23 @@ def test_cli_synthetic_rgb(tmp_path):
 
     spec = ProblemSpec(32, 24, 3, DType.U8, SHARPEN_5, 2)
     np.testing.assert_array_equal(got, run_reference(spec, img))
+
+
+def test_bench_cpu_world2(tmp_path):
+    """bench.py's distributed flow (torchrun world 2) on the CPU/gloo
+    backend — exercises the exact rank/barrier/allreduce path the 8-GPU
+    driver run takes, minus RCCL."""
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=2", "--master-addr=127.0.0.1",
+         "--master-port=29617", "bench.py", "--gpus", "2", "--steps", "3",
+         "--warmup", "1", "--preset", "single-8192", "--backend", "cpu"],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=600,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    rec = json.loads(line)
+    assert rec["n_gpus"] == 2 and rec["steps"] == 3
+    assert rec["metric"] == "Gpixel-iters/sec"
