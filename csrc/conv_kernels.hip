@@ -216,7 +216,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
     sq[k] = i % QPR;
   }
 
-  // prologue: stage tile 0
+  // prologue: stage tile 0 (synchronous load+convert+write)
 #pragma unroll
   for (int k = 0; k < NITER; ++k) {
     if (k == NITER - 1 && ITEMS % NTHREADS != 0 && tid >= ITEMS - k * NTHREADS)
@@ -226,7 +226,6 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
     Raw raw = TT::loadRaw(src + (size_t)gy * pitchE + ga);
     TT::cvtWrite(&lds[0][srow[k]][sq[k]], raw);
   }
-  __syncthreads();
 
   // compute lane mapping: one wave per output row-pair
   const int tx = tid & 63;
@@ -234,21 +233,28 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
   const int ex = etx0 + VEC * tx;
   const int c4 = 4 + tx - NL;  // first float4 col of this thread's window
 
-  for (int t = 0; t < nt; ++t) {
-    // issue next tile's global loads BEFORE the ds_reads of this tile
-    Raw pre[NITER];
-    const bool pf = t + 1 < nt;
-    if (pf) {
+  // issue the staging loads for tile `tile` into regs
+  auto issue = [&](Raw (&pre)[NITER], int tile) {
 #pragma unroll
-      for (int k = 0; k < NITER; ++k) {
-        if (k == NITER - 1 && ITEMS % NTHREADS != 0 && tid >= ITEMS - k * NTHREADS)
-      break;
-        const int gy = min(tyBase + (t + 1) * TILE_Y - R + srow[k], bufh - 1);
-        const int ga = max(0, min(gx0 + 4 * sq[k], pitchE - 4));
-        pre[k] = TT::loadRaw(src + (size_t)gy * pitchE + ga);
-      }
+    for (int k = 0; k < NITER; ++k) {
+      if (k == NITER - 1 && ITEMS % NTHREADS != 0 &&
+          tid >= ITEMS - k * NTHREADS)
+        break;
+      const int gy = min(tyBase + tile * TILE_Y - R + srow[k], bufh - 1);
+      const int ga = max(0, min(gx0 + 4 * sq[k], pitchE - 4));
+      pre[k] = TT::loadRaw(src + (size_t)gy * pitchE + ga);
     }
-
+  };
+  auto cvtW = [&](Raw (&pre)[NITER], int buf) {
+#pragma unroll
+    for (int k = 0; k < NITER; ++k) {
+      if (k == NITER - 1 && ITEMS % NTHREADS != 0 &&
+          tid >= ITEMS - k * NTHREADS)
+        break;
+      TT::cvtWrite(&lds[buf][srow[k]][sq[k]], pre[k]);
+    }
+  };
+  auto compute_store = [&](int t, int buf) {
     const int ty0 = tyBase + t * TILE_Y;
     float win[2 + 2 * R][NB * 4];
 #pragma unroll
@@ -258,7 +264,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
         // volatile LDS-addrspace vector load: stays one conflict-free
         // ds_read_b128 (plain loads get unused lanes DCE'd and split into
         // strided ds_read2_b32 -- 4-way bank conflicts).
-        f32x4 v = *(lds_v4*)&lds[t & 1][2 * tyy + rw][c4 + b];
+        f32x4 v = *(lds_v4*)&lds[buf][2 * tyy + rw][c4 + b];
         win[rw][4 * b] = v.x;
         win[rw][4 * b + 1] = v.y;
         win[rw][4 * b + 2] = v.z;
@@ -285,19 +291,6 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
                 fmaf(c, win[j + dy][4 * NL - SR + p + S * dx], acc[j][p]);
       }
 
-    // write-late: convert + LDS-write the prefetched tile BEFORE issuing
-    // this tile's global stores — the vmcnt wait here then only covers the
-    // prefetch loads (which had the whole compute phase to land), not the
-    // stores.
-    if (pf) {
-#pragma unroll
-      for (int k = 0; k < NITER; ++k) {
-        if (k == NITER - 1 && ITEMS % NTHREADS != 0 && tid >= ITEMS - k * NTHREADS)
-      break;
-        TT::cvtWrite(&lds[(t + 1) & 1][srow[k]][sq[k]], pre[k]);
-      }
-    }
-
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
       const int y = ty0 + 2 * tyy + j;
@@ -311,18 +304,34 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
           if (ex + p >= ex0 && ex + p < ex1) TT::store1(out + p, acc[j][p]);
       }
     }
+  };
 
-    // raw barrier with an LDS-only wait: __syncthreads() would also drain
-    // vmcnt(0), i.e. this tile's global STORES (CDNA4 vmcnt counts
-    // stores), serializing every tile behind its store latency
-    // (cdna_hip_programming.md §5 "the step-3 structure's ceiling").
-    // Only the LDS writes must be visible at the barrier.
+  // 2-deep prefetch pipeline: tile t+2's loads issue at the START of
+  // tile t, and are converted+written to LDS at the END of tile t+1 —
+  // every staging load gets two full compute phases to land (the 1-deep
+  // version left SQ_WAIT_ANY at ~46 % of wave cycles). The loop body
+  // covers two tiles so the two register generations keep static names
+  // (a runtime-indexed Raw array would spill to scratch).
+  Raw preA[NITER], preB[NITER];
+  if (nt > 1) issue(preA, 1);
+  __syncthreads();
+
+  for (int tt = 0; tt < nt; tt += 2) {
+    if (tt + 2 < nt) issue(preB, tt + 2);
+    compute_store(tt, 0);
+    if (tt + 1 < nt) cvtW(preA, 1);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (tt + 1 >= nt) break;
+    if (tt + 3 < nt) issue(preA, tt + 3);
+    compute_store(tt + 1, 1);
+    if (tt + 2 < nt) cvtW(preB, 0);
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
   }
 }
 
-// ---- MFMA path (BASELINE.json north-star: "MFMA fp16/fp32 accumulation") --
+// ---- MFMA path// ---- MFMA path (BASELINE.json north-star: "MFMA fp16/fp32 accumulation") --
 //
 // Formulation: for each filter row dy, the 1-D convolution along x of a
 // 16-row slice is a banded (Toeplitz) matrix product
