@@ -69,16 +69,26 @@ struct U8T {
   static __device__ __forceinline__ unsigned q1(float v) {
     return (unsigned)__float2int_rn(fminf(fmaxf(v, 0.f), 255.f));
   }
+  // v_cvt_pk_u8_f32 saturates to [0,255] and truncates; rndne first gives
+  // exactly the round-half-even + clamp contract in 2 ops/element instead
+  // of the rndne+clamp+cvt+shift-or chain (byte-exactness covered by the
+  // oracle tests: EDGE_3 drives negatives, sharpen5 overshoots 255).
+  static __device__ __forceinline__ unsigned pack4(const float* a) {
+    unsigned p = 0;
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      p = __builtin_amdgcn_cvt_pk_u8_f32(__builtin_rintf(a[i]), i, p);
+    return p;
+  }
   static __device__ __forceinline__ void store4(T* p, const float* a) {
-    *reinterpret_cast<unsigned*>(p) =
-        q1(a[0]) | (q1(a[1]) << 8) | (q1(a[2]) << 16) | (q1(a[3]) << 24);
+    *reinterpret_cast<unsigned*>(p) = pack4(a);
   }
   static __device__ __forceinline__ void store1(T* p, float v) { *p = (T)q1(v); }
   static __device__ __forceinline__ float toF(T v) { return (float)v; }
   static __device__ __forceinline__ void store8(T* p, const float* a) {
     uint2 u;
-    u.x = q1(a[0]) | (q1(a[1]) << 8) | (q1(a[2]) << 16) | (q1(a[3]) << 24);
-    u.y = q1(a[4]) | (q1(a[5]) << 8) | (q1(a[6]) << 16) | (q1(a[7]) << 24);
+    u.x = pack4(a);
+    u.y = pack4(a + 4);
     *reinterpret_cast<uint2*>(p) = u;
   }
   // fp16 staging for the MFMA path: 8 bytes -> 8 halves (u8 is exact in fp16)
