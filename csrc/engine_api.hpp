@@ -281,19 +281,6 @@ class ConvEngine {
                                hipMemcpyDeviceToDevice, comm_));
   }
 
-  void exchange(int d) {
-    // comm must see the latest src (written on compute)
-    HIP_CHECK(hipEventRecord(evCompute_, compute_));
-    HIP_CHECK(hipStreamWaitEvent(comm_, evCompute_, 0));
-    if (comm_nccl_) {
-      exchange_nccl(d);
-    } else {
-      exchange_local(d);
-    }
-    HIP_CHECK(hipEventRecord(evComm_, comm_));
-    HIP_CHECK(hipStreamWaitEvent(compute_, evComm_, 0));
-  }
-
   // LOCAL mode: neighbors are local block indices; D2D copies.
   void exchange_local(int d) {
     const int is = itemsize(dt_);
