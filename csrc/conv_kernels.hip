@@ -33,6 +33,7 @@
 #include <hip/hip_fp16.h>
 
 #include <algorithm>
+#include <cstdlib>
 
 namespace pconv {
 
@@ -592,6 +593,14 @@ __global__ __launch_bounds__(NTHREADS) void diff_max_k(
 
 // ---- dispatch -------------------------------------------------------------
 
+inline int chain_override() {
+  static int v = [] {
+    const char* e = getenv("PCONV_CHAIN");
+    return e ? atoi(e) : 0;
+  }();
+  return v;
+}
+
 template <typename TT, int R, int S>
 void launch_tile(const void* src, void* dst, const float* wgt,
                  const BlockGeom& g, const Region& r, hipStream_t stream) {
@@ -601,6 +610,7 @@ void launch_tile(const void* src, void* dst, const float* wgt,
   // chain length: keep >= ~4096 workgroups in flight (256 CUs x 8 blocks
   // with headroom); longer chains amortize the staging prologue.
   int chainLen = std::max(1, std::min(16, (tilesX * tilesY) / 4096));
+  if (chain_override() > 0) chainLen = chain_override();
   const int chains = (tilesY + chainLen - 1) / chainLen;
   hipLaunchKernelGGL((conv_tile_k<TT, R, S>), dim3(tilesX * chains),
                      dim3(NTHREADS), 0, stream,
@@ -617,6 +627,7 @@ void launch_mfma(const void* src, void* dst, const float* wgt,
   const int tilesX = (r.ex1 - abase + M_TILE_X - 1) / M_TILE_X;
   const int tilesY = (r.py1 - r.py0 + M_TILE_Y - 1) / M_TILE_Y;
   int chainLen = std::max(1, std::min(16, (tilesX * tilesY) / 4096));
+  if (chain_override() > 0) chainLen = chain_override();
   const int chains = (tilesY + chainLen - 1) / chainLen;
   hipLaunchKernelGGL((conv_mfma_k<TT, R, S>), dim3(tilesX * chains),
                      dim3(NTHREADS), 0, stream,

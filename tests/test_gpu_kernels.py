@@ -268,3 +268,58 @@ def test_mfma_multiblock_protocol():
         del os.environ["PCONV_FORCE_MFMA"]
     ref = run_reference(spec, img)
     np.testing.assert_array_equal(out, ref)
+
+
+def test_local_blocks_gpu_f32():
+    spec = ProblemSpec(128, 96, 1, DType.F32, GAUSSIAN_BLUR_3, 12)
+    img = synthetic_image(96, 128, 1, DType.F32)
+    out = _engine_mod().run_local_blocks_gpu(spec, img, 4)
+    ref = run_reference(spec, img)
+    np.testing.assert_allclose(out, ref, rtol=1e-5, atol=1e-3)
+
+
+def test_phase_timing_accumulates():
+    from parallel_convolution_amd.hip_engine import HipEngine
+    from parallel_convolution_amd.parallel.decomp import make_blocks
+
+    spec = ProblemSpec(256, 128, 1, DType.U8, GAUSSIAN_BLUR_3, 20)
+    img = synthetic_image(128, 256, 1, DType.U8)
+    blocks = make_blocks(128, 256, 2, 1)
+    eng = HipEngine(spec, blocks)
+    eng.set_phase_timing(True)
+    for i, bs in enumerate(blocks):
+        eng.load_block(i, img[bs.y0 : bs.y0 + bs.h, bs.x0 : bs.x0 + bs.w])
+    eng.run()
+    pt = eng.phase_times()
+    assert pt["chunks"] >= 1
+    assert pt["compute_ms"] > 0
+    assert pt["exchange_ms"] >= 0
+
+
+def test_cli_hip_checkpoint_resume(tmp_path):
+    import subprocess
+    import sys
+    import os
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    img = synthetic_image(96, 128, 1, DType.U8)
+    ip = str(tmp_path / "img.raw")
+    img.tofile(ip)
+    out = str(tmp_path / "out.raw")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+    def cmd(iters, extra):
+        return [sys.executable, "-m", "parallel_convolution_amd", ip, "128",
+                "96", "blur3", str(iters), "--backend", "hip", "--output",
+                out] + extra
+
+    r = subprocess.run(cmd(6, ["--checkpoint-every", "6"]), cwd=repo, env=env,
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr
+    r = subprocess.run(cmd(14, ["--checkpoint-every", "6", "--resume"]),
+                       cwd=repo, env=env, capture_output=True, text=True,
+                       timeout=300)
+    assert r.returncode == 0, r.stderr
+    got = np.fromfile(out, dtype=np.uint8).reshape(96, 128)
+    spec = ProblemSpec(128, 96, 1, DType.U8, GAUSSIAN_BLUR_3, 14)
+    np.testing.assert_array_equal(got, run_reference(spec, img))
