@@ -35,6 +35,7 @@
 
 #include <algorithm>
 #include <cstdlib>
+#include <unordered_map>
 
 namespace pconv {
 
@@ -188,7 +189,7 @@ template <typename TT, int R, int S>
 __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
     const typename TT::T* __restrict__ src, typename TT::T* __restrict__ dst,
     const float* __restrict__ wgt, int pitchE, int bufh, int ex0, int ex1,
-    int py0, int py1, int abase, int tilesX, int chainLen) {
+    int py0, int py1, int abase, int tilesX, int chainLen, int nChains) {
   using T = typename TT::T;
   using Raw = typename TT::Raw;
   constexpr int K = 2 * R + 1;
@@ -207,17 +208,24 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
   __shared__ f32x4 lds[2][ROWS_IN][LDS_P4];
 
   const int tid = threadIdx.x;
-  const int bx = blockIdx.x % tilesX;
-  const int bc = blockIdx.x / tilesX;
-  const int etx0 = abase + TILE_X * bx;  // tile origin (16-aligned element)
-  const int gx0 = etx0 - 16;             // staged window origin
-  const int tyBase = py0 + TILE_Y * chainLen * bc;
-  const int nt = min(chainLen, (py1 - tyBase + TILE_Y - 1) / TILE_Y);
 
   // filter -> registers (wave-uniform address -> scalar loads)
   float w[K * K];
 #pragma unroll
   for (int i = 0; i < K * K; ++i) w[i] = wgt[i];
+
+  // grid-stride over chain work items: an occupancy-sized grid avoids the
+  // tail quantization a one-WG-per-chain launch pays when the work count
+  // is not a multiple of the resident-WG count (25 % of wall at the
+  // 8-GPU block shape).
+  const int nWork = tilesX * nChains;
+  for (int work = blockIdx.x; work < nWork; work += gridDim.x) {
+  const int bx = work % tilesX;
+  const int bc = work / tilesX;
+  const int etx0 = abase + TILE_X * bx;  // tile origin (16-aligned element)
+  const int gx0 = etx0 - 16;             // staged window origin
+  const int tyBase = py0 + TILE_Y * chainLen * bc;
+  const int nt = min(chainLen, (py1 - tyBase + TILE_Y - 1) / TILE_Y);
 
   // staging role: item i covers row i/QPR, quarter i%QPR
   int srow[NITER], sq[NITER];
@@ -341,6 +349,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
   }
+  }  // work loop
 }
 
 // ---- MFMA path// ---- MFMA path (BASELINE.json north-star: "MFMA fp16/fp32 accumulation") --
@@ -381,7 +390,7 @@ template <typename TT, int R, int S>
 __global__ __launch_bounds__(NTHREADS) void conv_mfma_k(
     const typename TT::T* __restrict__ src, typename TT::T* __restrict__ dst,
     const float* __restrict__ wgt, int pitchE, int bufh, int ex0, int ex1,
-    int py0, int py1, int abase, int tilesX, int chainLen) {
+    int py0, int py1, int abase, int tilesX, int chainLen, int nChains) {
   using T = typename TT::T;
   using RawH = typename TT::RawH;
   constexpr int K = 2 * R + 1;
@@ -400,13 +409,6 @@ __global__ __launch_bounds__(NTHREADS) void conv_mfma_k(
   auto& lds = sm.img;
 
   const int tid = threadIdx.x;
-  const int bx = blockIdx.x % tilesX;
-  const int bc = blockIdx.x / tilesX;
-  const int etx0 = abase + M_TILE_X * bx;
-  const int ws = etx0 - 16;  // staged window origin (element)
-  const int tyBase = py0 + M_TILE_Y * chainLen * bc;
-  const int nt = min(chainLen, (py1 - tyBase + M_TILE_Y - 1) / M_TILE_Y);
-
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int jcol = lane & 15;
@@ -435,6 +437,17 @@ __global__ __launch_bounds__(NTHREADS) void conv_mfma_k(
     srow[k] = i / M_QPR;
     sq[k] = i % M_QPR;
   }
+
+  // grid-stride over chain work items (tail-quantization fix, see the
+  // VALU kernel note)
+  const int nWork = tilesX * nChains;
+  for (int work = blockIdx.x; work < nWork; work += gridDim.x) {
+  const int bx = work % tilesX;
+  const int bc = work / tilesX;
+  const int etx0 = abase + M_TILE_X * bx;
+  const int ws = etx0 - 16;  // staged window origin (element)
+  const int tyBase = py0 + M_TILE_Y * chainLen * bc;
+  const int nt = min(chainLen, (py1 - tyBase + M_TILE_Y - 1) / M_TILE_Y);
 
 #pragma unroll
   for (int k = 0; k < NITER; ++k) {
@@ -518,6 +531,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_mfma_k(
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
   }
+  }  // work loop
 }
 
 // A/B fragment-layout probe: D = A.B for one mfma_f32_16x16x32_f16 with
@@ -602,6 +616,34 @@ inline int chain_override() {
   return v;
 }
 
+// occupancy-sized grid: one pass of resident workgroups grid-strides the
+// chain work items, so an awkward work count costs a remainder of TILES,
+// not a whole extra wave of chains
+inline int grid_for(const void* kernel, int nWork) {
+  static thread_local std::unordered_map<const void*, int> cache;
+  auto it = cache.find(kernel);
+  int perCU;
+  if (it != cache.end()) {
+    perCU = it->second;
+  } else {
+    perCU = 1;
+    (void)hipOccupancyMaxActiveBlocksPerMultiprocessor(&perCU, kernel,
+                                                       NTHREADS, 0);
+    if (perCU < 1) perCU = 1;
+    cache[kernel] = perCU;
+  }
+  int dev = 0;
+  (void)hipGetDevice(&dev);
+  static thread_local int numCU = 0;
+  if (numCU == 0) {
+    hipDeviceProp_t prop;
+    if (hipGetDeviceProperties(&prop, dev) == hipSuccess)
+      numCU = prop.multiProcessorCount;
+    if (numCU <= 0) numCU = 256;
+  }
+  return std::max(1, std::min(nWork, perCU * numCU));
+}
+
 template <typename TT, int R, int S>
 void launch_tile(const void* src, void* dst, const float* wgt,
                  const BlockGeom& g, const Region& r, hipStream_t stream) {
@@ -613,12 +655,13 @@ void launch_tile(const void* src, void* dst, const float* wgt,
   int chainLen = std::max(1, std::min(16, (tilesX * tilesY) / 4096));
   if (chain_override() > 0) chainLen = chain_override();
   const int chains = (tilesY + chainLen - 1) / chainLen;
-  hipLaunchKernelGGL((conv_tile_k<TT, R, S>), dim3(tilesX * chains),
-                     dim3(NTHREADS), 0, stream,
-                     reinterpret_cast<const typename TT::T*>(src),
+  const int grid =
+      grid_for((const void*)&conv_tile_k<TT, R, S>, tilesX * chains);
+  hipLaunchKernelGGL((conv_tile_k<TT, R, S>), dim3(grid), dim3(NTHREADS), 0,
+                     stream, reinterpret_cast<const typename TT::T*>(src),
                      reinterpret_cast<typename TT::T*>(dst), wgt, g.pitchE,
                      g.bufh, r.ex0, r.ex1, r.py0, r.py1, abase, tilesX,
-                     chainLen);
+                     chainLen, chains);
 }
 
 template <typename TT, int R, int S>
@@ -630,12 +673,13 @@ void launch_mfma(const void* src, void* dst, const float* wgt,
   int chainLen = std::max(1, std::min(16, (tilesX * tilesY) / 4096));
   if (chain_override() > 0) chainLen = chain_override();
   const int chains = (tilesY + chainLen - 1) / chainLen;
-  hipLaunchKernelGGL((conv_mfma_k<TT, R, S>), dim3(tilesX * chains),
-                     dim3(NTHREADS), 0, stream,
-                     reinterpret_cast<const typename TT::T*>(src),
+  const int grid =
+      grid_for((const void*)&conv_mfma_k<TT, R, S>, tilesX * chains);
+  hipLaunchKernelGGL((conv_mfma_k<TT, R, S>), dim3(grid), dim3(NTHREADS), 0,
+                     stream, reinterpret_cast<const typename TT::T*>(src),
                      reinterpret_cast<typename TT::T*>(dst), wgt, g.pitchE,
                      g.bufh, r.ex0, r.ex1, r.py0, r.py1, abase, tilesX,
-                     chainLen);
+                     chainLen, chains);
 }
 
 template <typename TT>
