@@ -657,6 +657,10 @@ void launch_tile(const void* src, void* dst, const float* wgt,
   const int chains = (tilesY + chainLen - 1) / chainLen;
   const int grid =
       grid_for((const void*)&conv_tile_k<TT, R, S>, tilesX * chains);
+  if (getenv("PCONV_DEBUG_GRID"))
+    fprintf(stderr, "[launch] tilesX=%d tilesY=%d chainLen=%d chains=%d nWork=%d grid=%d region=[%d,%d)x[%d,%d)\n",
+            tilesX, tilesY, chainLen, chains, tilesX * chains, grid, r.py0,
+            r.py1, r.ex0, r.ex1);
   hipLaunchKernelGGL((conv_tile_k<TT, R, S>), dim3(grid), dim3(NTHREADS), 0,
                      stream, reinterpret_cast<const typename TT::T*>(src),
                      reinterpret_cast<typename TT::T*>(dst), wgt, g.pitchE,

@@ -159,6 +159,12 @@ class HipEngine:
         iters = self.spec.iters if iters is None else iters
         if iters <= 0:
             return
+        # The engine runs on its own HIP streams: torch-side work (the
+        # buffer zero-fill at construction, load_block's H2D copies) must
+        # complete first or it races the conv kernels — at 65536² the
+        # 8.6 GB fill demonstrably overwrote conv output (see
+        # bench/validate_large.py, which caught this).
+        torch.cuda.synchronize()
         self.engine.run(iters, self.chunk_iters)
         self._iters_done += iters
 
@@ -166,6 +172,7 @@ class HipEngine:
         """max |last - previous| over owned regions of local blocks (the
         convergence test of SURVEY.md §2.1 R10; allreduce is the caller's
         job in distributed mode)."""
+        torch.cuda.synchronize()
         return float(self.engine.diff_max())
 
     def set_phase_timing(self, on: bool) -> None:
