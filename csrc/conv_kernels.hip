@@ -384,9 +384,10 @@ constexpr int M_TILE_X = 256;  // output elements per workgroup tile
 constexpr int M_TILE_Y = 16;
 constexpr int M_P2 = 320;  // f16 per LDS row (640 B; XOR swizzle stays in-row)
 constexpr int M_QPR = 36;  // staged 16-B quarters per row (288-element window)
+constexpr int M_NT = 256;  // MFMA kernel workgroup (4 waves of 16x16 subtile quads)
 
 template <typename TT, int R, int S>
-__global__ __launch_bounds__(NTHREADS) void conv_mfma_k(
+__global__ __launch_bounds__(M_NT) void conv_mfma_k(
     const typename TT::T* __restrict__ src, typename TT::T* __restrict__ dst,
     const float* __restrict__ wgt, int pitchE, int bufh, int ex0, int ex1,
     int py0, int py1, int abase, int tilesX, int chainLen, int nChains) {
@@ -395,7 +396,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_mfma_k(
   constexpr int K = 2 * R + 1;
   constexpr int ROWS_IN = M_TILE_Y + 2 * R;
   constexpr int ITEMS = ROWS_IN * M_QPR;
-  constexpr int NITER = (ITEMS + NTHREADS - 1) / NTHREADS;
+  constexpr int NITER = (ITEMS + M_NT - 1) / M_NT;
   // C-transpose scratch row pitch (bytes): 64 elements + pad chosen so the
   // 16-B read-back slots of one lane group land on distinct banks
   constexpr int CSPB = sizeof(T) == 1 ? 112 : 64 * sizeof(T) + 16;
@@ -432,7 +433,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_mfma_k(
   int srow[NITER], sq[NITER];
 #pragma unroll
   for (int k = 0; k < NITER; ++k) {
-    const int i = tid + k * NTHREADS;
+    const int i = tid + k * M_NT;
     srow[k] = i / M_QPR;
     sq[k] = i % M_QPR;
   }
@@ -450,7 +451,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_mfma_k(
 
 #pragma unroll
   for (int k = 0; k < NITER; ++k) {
-    if (k == NITER - 1 && ITEMS % NTHREADS != 0 && tid >= ITEMS - k * NTHREADS)
+    if (k == NITER - 1 && ITEMS % M_NT != 0 && tid >= ITEMS - k * M_NT)
       break;
     const int gy = min(max(tyBase - R + srow[k], 0), bufh - 1);
     const int ga = max(0, min(ws + 8 * sq[k], pitchE - 8));
@@ -465,7 +466,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_mfma_k(
     if (pf) {
 #pragma unroll
       for (int k = 0; k < NITER; ++k) {
-        if (k == NITER - 1 && ITEMS % NTHREADS != 0 && tid >= ITEMS - k * NTHREADS)
+        if (k == NITER - 1 && ITEMS % M_NT != 0 && tid >= ITEMS - k * M_NT)
       break;
         const int gy = min(tyBase + (t + 1) * M_TILE_Y - R + srow[k], bufh - 1);
         const int ga = max(0, min(ws + 8 * sq[k], pitchE - 8));
@@ -521,7 +522,7 @@ __global__ __launch_bounds__(NTHREADS) void conv_mfma_k(
     if (pf) {
 #pragma unroll
       for (int k = 0; k < NITER; ++k) {
-        if (k == NITER - 1 && ITEMS % NTHREADS != 0 && tid >= ITEMS - k * NTHREADS)
+        if (k == NITER - 1 && ITEMS % M_NT != 0 && tid >= ITEMS - k * M_NT)
       break;
         TT::cvtWriteH(&lds[(t + 1) & 1][srow[k]][(sq[k] ^ (srow[k] & 7)) * 8],
                       pre[k]);
@@ -618,7 +619,7 @@ inline int chain_override() {
 // occupancy-sized grid: one pass of resident workgroups grid-strides the
 // chain work items, so an awkward work count costs a remainder of TILES,
 // not a whole extra wave of chains
-inline int grid_for(const void* kernel, int nWork) {
+inline int grid_for(const void* kernel, int nWork, int nThreads = NTHREADS) {
   static thread_local std::unordered_map<const void*, int> cache;
   auto it = cache.find(kernel);
   int perCU;
@@ -627,7 +628,7 @@ inline int grid_for(const void* kernel, int nWork) {
   } else {
     perCU = 1;
     (void)hipOccupancyMaxActiveBlocksPerMultiprocessor(&perCU, kernel,
-                                                       NTHREADS, 0);
+                                                       nThreads, 0);
     if (perCU < 1) perCU = 1;
     cache[kernel] = perCU;
   }
@@ -677,8 +678,8 @@ void launch_mfma(const void* src, void* dst, const float* wgt,
   if (chain_override() > 0) chainLen = chain_override();
   const int chains = (tilesY + chainLen - 1) / chainLen;
   const int grid =
-      grid_for((const void*)&conv_mfma_k<TT, R, S>, tilesX * chains);
-  hipLaunchKernelGGL((conv_mfma_k<TT, R, S>), dim3(grid), dim3(NTHREADS), 0,
+      grid_for((const void*)&conv_mfma_k<TT, R, S>, tilesX * chains, M_NT);
+  hipLaunchKernelGGL((conv_mfma_k<TT, R, S>), dim3(grid), dim3(M_NT), 0,
                      stream, reinterpret_cast<const typename TT::T*>(src),
                      reinterpret_cast<typename TT::T*>(dst), wgt, g.pitchE,
                      g.bufh, r.ex0, r.ex1, r.py0, r.py1, abase, tilesX,
