@@ -31,7 +31,7 @@ from parallel_convolution_amd.models.filters import (  # noqa: E402
     SHARPEN_5,
 )
 from parallel_convolution_amd.parallel import bootstrap  # noqa: E402
-from parallel_convolution_amd.parallel.decomp import make_blocks  # noqa: E402
+from parallel_convolution_amd.parallel.decomp import halo_for_radius, make_blocks  # noqa: E402
 from parallel_convolution_amd.spec import DType, ProblemSpec  # noqa: E402
 from parallel_convolution_amd.utils.rawio import synthetic_image  # noqa: E402
 
@@ -77,7 +77,7 @@ def main() -> int:
     if world > 1:
         bootstrap.init_control_group("gloo")
 
-    blocks = make_blocks(spec.height, spec.width, n_gpus, spec.channels)
+    blocks = make_blocks(spec.height, spec.width, n_gpus, spec.channels, halo=halo_for_radius(spec.radius))
     if world > 1:
         my_blocks = [blocks[rank]]
     else:

@@ -28,7 +28,7 @@ import numpy as np
 
 from .models.filters import read_filter
 from .parallel import bootstrap
-from .parallel.decomp import make_blocks
+from .parallel.decomp import halo_for_radius, make_blocks
 from .spec import DType, ProblemSpec
 from .utils import metrics, rawio
 
@@ -109,7 +109,7 @@ def run_rank(args) -> int:
     if backend == "hip":
         torch.cuda.set_device(local_rank)
 
-    blocks = make_blocks(spec.height, spec.width, world, spec.channels)
+    blocks = make_blocks(spec.height, spec.width, world, spec.channels, halo=halo_for_radius(spec.radius))
     my = blocks[rank]
     S = spec.channels
 

@@ -26,6 +26,7 @@ import torch.nn.functional as F
 from .parallel.decomp import (
     BlockSpec,
     ChunkPlan,
+    halo_for_radius,
     make_blocks,
     max_chunk_iters,
     plan_chunks,
@@ -260,7 +261,9 @@ class CpuEngine:
 
 def run_single_cpu(spec: ProblemSpec, image: np.ndarray) -> np.ndarray:
     """Convenience: whole image, 1 rank, CPU (BASELINE.json:7 config 1)."""
-    blocks = make_blocks(spec.height, spec.width, 1, spec.channels)
+    blocks = make_blocks(
+        spec.height, spec.width, 1, spec.channels, halo=halo_for_radius(spec.radius)
+    )
     blk = CpuBlock(spec, blocks[0])
     blk.load(image)
     eng = CpuEngine(spec, [blk], exchanger=None)
@@ -270,7 +273,9 @@ def run_single_cpu(spec: ProblemSpec, image: np.ndarray) -> np.ndarray:
 
 def run_local_blocks(spec: ProblemSpec, image: np.ndarray, p: int) -> np.ndarray:
     """P blocks in one process with LocalExchanger — protocol test path."""
-    specs = make_blocks(spec.height, spec.width, p, spec.channels)
+    specs = make_blocks(
+        spec.height, spec.width, p, spec.channels, halo=halo_for_radius(spec.radius)
+    )
     blocks = []
     for bs in specs:
         blk = CpuBlock(spec, bs)

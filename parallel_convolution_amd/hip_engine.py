@@ -18,7 +18,7 @@ from typing import List, Optional
 import numpy as np
 import torch
 
-from .parallel.decomp import BlockSpec, make_blocks, max_chunk_iters
+from .parallel.decomp import BlockSpec, halo_for_radius, make_blocks, max_chunk_iters
 from .spec import DType, ProblemSpec
 
 _DT_CODE = {DType.U8: 0, DType.F16: 1, DType.F32: 2}
@@ -190,7 +190,9 @@ class HipEngine:
 
 def run_single_gpu(spec: ProblemSpec, image: np.ndarray, device: str = "cuda") -> np.ndarray:
     """Whole image on one GPU (BASELINE.json:8 config 2 path)."""
-    blocks = make_blocks(spec.height, spec.width, 1, spec.channels)
+    blocks = make_blocks(
+        spec.height, spec.width, 1, spec.channels, halo=halo_for_radius(spec.radius)
+    )
     eng = HipEngine(spec, blocks)
     eng.load_block(0, image)
     eng.run()
@@ -200,7 +202,9 @@ def run_single_gpu(spec: ProblemSpec, image: np.ndarray, device: str = "cuda") -
 def run_local_blocks_gpu(spec: ProblemSpec, image: np.ndarray, p: int) -> np.ndarray:
     """P blocks on ONE GPU with D2D exchange — the 1-GPU-box test mode for
     the multi-GPU protocol."""
-    specs = make_blocks(spec.height, spec.width, p, spec.channels)
+    specs = make_blocks(
+        spec.height, spec.width, p, spec.channels, halo=halo_for_radius(spec.radius)
+    )
     eng = HipEngine(spec, specs)
     S = spec.channels
     for i, bs in enumerate(specs):

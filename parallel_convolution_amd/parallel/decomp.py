@@ -31,9 +31,17 @@ from __future__ import annotations
 import dataclasses
 from typing import Dict, List, Optional, Tuple
 
-#: Allocated halo width (pixels). Also the default max exchange depth.
-#: 16 keeps tile x-origins 16-byte aligned for the u8 kernels.
+#: Default allocated halo width (pixels). Also the default max exchange
+#: depth. 16 keeps tile x-origins 16-byte aligned for the u8 kernels.
 HALO_CAP = 16
+
+
+def halo_for_radius(radius: int) -> int:
+    """Allocated halo width for a filter radius: at least HALO_CAP, and at
+    least the radius itself (rounded up to 16 to keep alignment), so large
+    odd kernels (k > 33) never read past the padded buffer."""
+    need = max(HALO_CAP, radius)
+    return (need + 15) // 16 * 16
 
 #: Row pitch alignment in storage ELEMENTS (64 elems → 64/128/256 B for
 #: u8/f16/f32 rows: every row start is at least 64-byte aligned).

@@ -106,3 +106,16 @@ def test_random_float_filter_tolerance():
     diff = np.abs(out.astype(np.int32) - ref.astype(np.int32))
     assert diff.max() <= 1
     assert (diff > 0).mean() < 0.01
+
+
+def test_large_kernel_k35():
+    """k=35 (radius 17 > default halo): halo_for_radius must widen the
+    padding so nothing reads past the buffer."""
+    rng = np.random.default_rng(5)
+    filt = rng.random((35, 35), dtype=np.float32)
+    filt /= filt.sum()
+    spec = ProblemSpec(96, 80, 1, DType.F32, filt, 2)
+    img = synthetic_image(80, 96, 1, DType.F32)
+    out = run_local_blocks(spec, img, 2)
+    ref = run_reference(spec, img)
+    np.testing.assert_allclose(out, ref, rtol=1e-4, atol=1e-2)

@@ -323,3 +323,16 @@ def test_cli_hip_checkpoint_resume(tmp_path):
     got = np.fromfile(out, dtype=np.uint8).reshape(96, 128)
     spec = ProblemSpec(128, 96, 1, DType.U8, GAUSSIAN_BLUR_3, 14)
     np.testing.assert_array_equal(got, run_reference(spec, img))
+
+
+def test_generic_kernel_k35_gpu():
+    """Generic-kernel fallback with radius 17 (> default halo): the halo
+    must widen so the kernel never reads outside the padded buffer."""
+    rng = np.random.default_rng(5)
+    filt = rng.random((35, 35), dtype=np.float32)
+    filt /= filt.sum()
+    spec = ProblemSpec(128, 96, 1, DType.F32, filt, 2)
+    img = synthetic_image(96, 128, 1, DType.F32)
+    out = _engine_mod().run_single_gpu(spec, img)
+    ref = run_reference(spec, img)
+    np.testing.assert_allclose(out, ref, rtol=1e-4, atol=1e-2)
