@@ -29,9 +29,7 @@
 #include <rccl/rccl.h>
 
 #include <algorithm>
-#include <array>
 #include <cstring>
-#include <memory>
 
 #define NCCL_CHECK(expr)                                                      \
   do {                                                                        \
@@ -116,7 +114,6 @@ class ConvEngine {
     if (uid.size() < sizeof(id)) throw std::runtime_error("bad nccl uid size");
     std::memcpy(&id, uid.data(), sizeof(id));
     NCCL_CHECK(ncclCommInitRank(&comm_nccl_, world, id, rank));
-    rank_ = rank;
   }
 
   static std::string get_unique_id() {
@@ -372,7 +369,6 @@ class ConvEngine {
   DType dt_;
   int k_;
   int S_;
-  int rank_ = 0;
   std::vector<DeviceBlock> blocks_;
   hipStream_t compute_ = nullptr;
   hipStream_t comm_ = nullptr;

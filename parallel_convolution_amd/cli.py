@@ -22,9 +22,6 @@ import signal
 import subprocess
 import sys
 import time
-from typing import Optional
-
-import numpy as np
 
 from .models.filters import read_filter
 from .parallel import bootstrap

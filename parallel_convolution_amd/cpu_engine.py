@@ -25,7 +25,6 @@ import torch.nn.functional as F
 
 from .parallel.decomp import (
     BlockSpec,
-    ChunkPlan,
     halo_for_radius,
     make_blocks,
     max_chunk_iters,
@@ -207,10 +206,6 @@ class RunStats:
     iters: int
     elapsed_s: float
     exchanges: int
-
-    @property
-    def pixel_iters(self) -> int:
-        return 0  # filled by caller
 
 
 class CpuEngine:

@@ -29,7 +29,7 @@ block lives at [halo:halo+h, halo:halo+w] (pixel coords; multiply x by
 from __future__ import annotations
 
 import dataclasses
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 #: Default allocated halo width (pixels). Also the default max exchange
 #: depth. 16 keeps tile x-origins 16-byte aligned for the u8 kernels.
