@@ -46,6 +46,8 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--gpus", type=int, default=1, help="GPUs (1 process per GPU)")
     p.add_argument("--backend", choices=["auto", "hip", "cpu"], default="auto")
     p.add_argument("--chunk", type=int, default=None, help="halo-deepening chunk iterations")
+    p.add_argument("--grid", default=None, metavar="RxC",
+                   help="process-grid override (default: near-square factorization)")
     p.add_argument("--seed", type=int, default=1234, help="seed for synthetic input")
     p.add_argument("--converge-every", type=int, default=0, metavar="N",
                    help="check max|Δ| every N iters, stop early below eps")
@@ -106,7 +108,18 @@ def run_rank(args) -> int:
     if backend == "hip":
         torch.cuda.set_device(local_rank)
 
-    blocks = make_blocks(spec.height, spec.width, world, spec.channels, halo=halo_for_radius(spec.radius))
+    grid = None
+    if args.grid:
+        gr, gc = args.grid.lower().split("x")
+        grid = (int(gr), int(gc))
+    blocks = make_blocks(
+        spec.height,
+        spec.width,
+        world,
+        spec.channels,
+        halo=halo_for_radius(spec.radius),
+        grid=grid,
+    )
     my = blocks[rank]
     S = spec.channels
 

@@ -136,3 +136,17 @@ def test_bench_cpu_world2(tmp_path):
     rec = json.loads(line)
     assert rec["n_gpus"] == 2 and rec["steps"] == 3
     assert rec["metric"] == "Gpixel-iters/sec"
+
+
+def test_cli_grid_override(image_and_filter, tmp_path):
+    ip, fp, img = image_and_filter
+    out = str(tmp_path / "outg.raw")
+    mpath = str(tmp_path / "mg.jsonl")
+    r = run_cli([ip, "64", "48", fp, "5", "--backend", "cpu", "--gpus", "2",
+                 "--grid", "2x1", "--output", out, "--metrics-out", mpath])
+    assert r.returncode == 0, r.stderr + r.stdout
+    rec = json.loads(open(mpath).read().strip())
+    assert rec["grid"] == "2x1"  # override actually applied (default is 1x2)
+    got = rawio.read_image(out, 48, 64, 1, DType.U8)
+    spec = ProblemSpec(64, 48, 1, DType.U8, GAUSSIAN_BLUR_3, 5)
+    np.testing.assert_array_equal(got, run_reference(spec, img))
