@@ -336,3 +336,31 @@ def test_generic_kernel_k35_gpu():
     out = _engine_mod().run_single_gpu(spec, img)
     ref = run_reference(spec, img)
     np.testing.assert_allclose(out, ref, rtol=1e-4, atol=1e-2)
+
+
+@pytest.mark.parametrize("extra,dt,C", [
+    (["--dtype", "fp16"], DType.F16, 1),
+    (["--rgb"], DType.U8, 3),
+])
+def test_cli_hip_dtype_variants(tmp_path, extra, dt, C):
+    import subprocess
+    import sys
+    import os
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    img = synthetic_image(48, 64, C, dt)
+    ip = str(tmp_path / "img.raw")
+    img.tofile(ip)
+    out = str(tmp_path / "out.raw")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+    r = subprocess.run(
+        [sys.executable, "-m", "parallel_convolution_amd", ip, "64", "48",
+         "blur3", "6", "--backend", "hip", "--output", out] + extra,
+        cwd=repo, env=env, capture_output=True, text=True, timeout=300,
+    )
+    assert r.returncode == 0, r.stderr
+    got = np.fromfile(out, dtype=dt.numpy_dtype).reshape(48, 64 * C)
+    spec = ProblemSpec(64, 48, C, dt, GAUSSIAN_BLUR_3, 6)
+    ref = run_reference(spec, img)
+    np.testing.assert_array_equal(got, ref)
