@@ -61,6 +61,13 @@ void launch_conv(DType dt, int S, int k, const void* src, void* dst,
                  const float* wgt, const BlockGeom& g, const Region& r,
                  bool useMfma, hipStream_t stream);
 
+// Fused double iteration: iteration A into an LDS mid image (quantized),
+// iteration B to global — rm is iteration A's region, r the output
+// region. Returns false if no fused template exists for (k, S).
+bool launch_conv2(DType dt, int S, int k, const void* src, void* dst,
+                  const float* wgt, const BlockGeom& g, const Region& rm,
+                  const Region& ro, hipStream_t stream);
+
 // one v_mfma_f32_16x16x32_f16 with the conv kernel's assumed fragment
 // layout: D[16][16] = A[16][32] . B[32][16] (layout verification probe)
 void launch_mfma_probe(const float* A32, const float* B32, float* D,

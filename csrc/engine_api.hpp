@@ -29,6 +29,7 @@
 #include <rccl/rccl.h>
 
 #include <algorithm>
+#include <cstdlib>
 #include <cstring>
 
 #define NCCL_CHECK(expr)                                                      \
@@ -70,6 +71,16 @@ class ConvEngine {
     HIP_CHECK(hipMemcpy(wgt_, filter.data(), filter.size() * sizeof(float),
                         hipMemcpyHostToDevice));
     HIP_CHECK(hipMalloc(&diffDev_, sizeof(float)));
+    // Fused double-iteration policy: halves HBM traffic per iteration —
+    // on by default for the bandwidth-bound fp16/fp32 storage paths
+    // (the u8 path is issue-bound; measured A/B in profiles/).
+    // PCONV_FUSE=0/1 overrides.
+    const char* fe = getenv("PCONV_FUSE");
+    if (fe) {
+      fusePairs_ = atoi(fe) != 0;
+    } else {
+      fusePairs_ = dt_ != DType::U8;
+    }
   }
 
   ~ConvEngine() {
@@ -179,7 +190,24 @@ class ConvEngine {
       if (phaseTiming_ && t0 == 0)
         HIP_CHECK(hipEventRecord(evPhase_[1], comm_));  // no exchange
       if (phaseTiming_) HIP_CHECK(hipEventRecord(evPhase_[2], compute_));
-      for (int t = t0; t < T; ++t) {
+      const bool canFuse =
+          fusePairs_ && !useMfma_ && r >= 1 && r <= 2 && (S_ == 1 || S_ == 3);
+      for (int t = t0; t < T;) {
+        if (canFuse && t + 1 < T) {
+          // fused pair: iteration t into LDS mid, iteration t+1 to global
+          bool ok = true;
+          for (auto& b : blocks_) {
+            ok &= launch_conv2(dt_, S_, k_, b.src, b.dst, (const float*)wgt_,
+                               b.g, region_for(b.g, d - (t + 1) * r),
+                               region_for(b.g, d - (t + 2) * r), compute_);
+          }
+          if (ok) {
+            for (auto& b : blocks_) std::swap(b.src, b.dst);
+            swapped_ = !swapped_;
+            t += 2;
+            continue;
+          }
+        }
         const int ext = d - (t + 1) * r;
         for (auto& b : blocks_) {
           launch_conv(dt_, S_, k_, b.src, b.dst, (const float*)wgt_, b.g,
@@ -187,6 +215,7 @@ class ConvEngine {
         }
         for (auto& b : blocks_) std::swap(b.src, b.dst);
         swapped_ = !swapped_;
+        ++t;
       }
       left -= T;
       if (phaseTiming_) {
@@ -379,6 +408,7 @@ class ConvEngine {
   ncclComm_t comm_nccl_ = nullptr;
   bool swapped_ = false;
   bool useMfma_ = false;
+  bool fusePairs_ = false;
   bool phaseTiming_ = false;
   hipEvent_t evPhase_[4] = {};
   double exchangeMs_ = 0.0;

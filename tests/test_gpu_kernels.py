@@ -364,3 +364,36 @@ def test_cli_hip_dtype_variants(tmp_path, extra, dt, C):
     spec = ProblemSpec(64, 48, C, dt, GAUSSIAN_BLUR_3, 6)
     ref = run_reference(spec, img)
     np.testing.assert_array_equal(got, ref)
+
+
+@pytest.mark.parametrize("p", [1, 4])
+def test_fused_double_iteration_byte_exact(p):
+    """Fused two-iterations-per-launch path (PCONV_FUSE=1) must agree
+    bitwise with the oracle for u8 (mid image quantized per iteration)."""
+    import os
+
+    spec = ProblemSpec(96, 80, 1, DType.U8, GAUSSIAN_BLUR_3, 21)  # odd iters
+    img = synthetic_image(80, 96, 1, DType.U8)
+    os.environ["PCONV_FUSE"] = "1"
+    try:
+        out = _engine_mod().run_local_blocks_gpu(spec, img, p)
+    finally:
+        del os.environ["PCONV_FUSE"]
+    ref = run_reference(spec, img)
+    np.testing.assert_array_equal(out, ref)
+
+
+def test_fused_rgb_k5():
+    import os
+
+    spec = ProblemSpec(128, 96, 3, DType.U8, SHARPEN_5, 7)
+    img = synthetic_image(96, 128, 3, DType.U8)
+    os.environ["PCONV_FUSE"] = "1"
+    os.environ["PCONV_DISABLE_MFMA"] = "1"
+    try:
+        out = _engine_mod().run_local_blocks_gpu(spec, img, 4)
+    finally:
+        del os.environ["PCONV_FUSE"]
+        del os.environ["PCONV_DISABLE_MFMA"]
+    ref = run_reference(spec, img)
+    np.testing.assert_array_equal(out, ref)
