@@ -41,6 +41,7 @@ PRESETS = {
     "rgb5": dict(width=16384, height=16384, channels=3, dtype="u8", filt="sharpen5"),
     "fp16-65536": dict(width=65536, height=65536, channels=1, dtype="fp16", filt="blur3"),
     "single-8192": dict(width=8192, height=8192, channels=1, dtype="u8", filt="blur3"),
+    "fp32-16384": dict(width=16384, height=16384, channels=1, dtype="fp32", filt="blur3"),
 }
 
 
