@@ -397,3 +397,31 @@ def test_fused_rgb_k5():
         del os.environ["PCONV_DISABLE_MFMA"]
     ref = run_reference(spec, img)
     np.testing.assert_array_equal(out, ref)
+
+
+def test_fused_f16_tie_rounding_consistency():
+    """fp16 exact-tie regression: the fused path's LDS quantizer must
+    round exactly like the store path (packed v_cvt_pk_f16_f32 == torch
+    round-half-even). Inputs are crafted so iteration 1 produces exact
+    fp16 ties: w = [.5, .5] row filter over alternating values whose mean
+    sits exactly between two fp16 neighbors."""
+    import os
+
+    filt = np.zeros((3, 3), dtype=np.float32)
+    filt[1, 1] = 0.5
+    filt[1, 2] = 0.5
+    a, b = 1.0, 1.0 + 3 * 2.0**-10  # mean = 1 + 1.5*2^-10: exact f16 tie
+    img = np.empty((64, 64), dtype=np.float16)
+    img[:, 0::2] = a
+    img[:, 1::2] = b
+    spec = ProblemSpec(64, 64, 1, DType.F16, filt, 2)
+    ref = run_reference(spec, img)
+    outs = {}
+    for fuse in ("0", "1"):
+        os.environ["PCONV_FUSE"] = fuse
+        try:
+            outs[fuse] = _engine_mod().run_single_gpu(spec, img)
+        finally:
+            del os.environ["PCONV_FUSE"]
+    np.testing.assert_array_equal(outs["0"], outs["1"])
+    np.testing.assert_array_equal(outs["0"], ref)
