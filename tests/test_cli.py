@@ -150,3 +150,36 @@ def test_cli_grid_override(image_and_filter, tmp_path):
     got = rawio.read_image(out, 48, 64, 1, DType.U8)
     spec = ProblemSpec(64, 48, 1, DType.U8, GAUSSIAN_BLUR_3, 5)
     np.testing.assert_array_equal(got, run_reference(spec, img))
+
+
+def test_cli_wrong_image_size_fails_cleanly(image_and_filter, tmp_path):
+    ip, fp, img = image_and_filter  # 64x48 image
+    r = run_cli([ip, "128", "128", fp, "2", "--backend", "cpu",
+                 "--output", str(tmp_path / "x.raw")])
+    assert r.returncode != 0
+    assert "short read" in r.stderr or "expected" in r.stderr
+
+
+def test_cli_missing_filter_file_fails(image_and_filter, tmp_path):
+    ip, _, _ = image_and_filter
+    r = run_cli([ip, "64", "48", str(tmp_path / "nope.txt"), "2",
+                 "--backend", "cpu", "--no-output"])
+    assert r.returncode != 0
+
+
+def test_cli_bad_filter_contents(image_and_filter, tmp_path):
+    ip, _, _ = image_and_filter
+    bad = tmp_path / "bad.txt"
+    bad.write_text("4\n1 2 3 4\n")  # even k
+    r = run_cli([ip, "64", "48", str(bad), "2", "--backend", "cpu",
+                 "--no-output"])
+    assert r.returncode != 0
+    assert "odd" in r.stderr
+
+
+def test_cli_watchdog_smoke(image_and_filter, tmp_path):
+    ip, fp, img = image_and_filter
+    # generous watchdog: must not fire on a normal run
+    r = run_cli([ip, "64", "48", fp, "3", "--backend", "cpu",
+                 "--watchdog", "120", "--no-output"])
+    assert r.returncode == 0, r.stderr
