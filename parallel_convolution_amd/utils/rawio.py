@@ -36,6 +36,19 @@ def write_image(path: str, img: np.ndarray) -> None:
     np.ascontiguousarray(img).tofile(path)
 
 
+def _native_io():
+    """The C++ block-I/O (csrc/io.cpp) when the extension is built; the
+    Python fallback keeps this module importable without it."""
+    try:
+        import torch  # noqa: F401 — extension links against torch
+
+        from .. import _C
+
+        return _C
+    except ImportError:
+        return None
+
+
 def read_block(
     path: str,
     height: int,
@@ -49,11 +62,19 @@ def read_block(
 ) -> np.ndarray:
     """Read one rank's (h, w) pixel block: returns (h, w*channels).
 
-    Seeks to each row's offset like the reference's per-rank fread+fseek
-    (SURVEY.md §2.3 'input scatter' row).
+    Per-rank row-offset reads like the reference's fread+fseek scatter
+    (SURVEY.md §2.3 'input scatter' row); served by the native pread
+    implementation (csrc/io.cpp) when the extension is available.
     """
     row_elems = width * channels
     itemsize = dtype.itemsize
+    C = _native_io()
+    if C is not None:
+        import torch
+
+        out_t = torch.empty((h, w * channels), dtype=dtype.torch_dtype)
+        C.read_block(path, row_elems * itemsize, x0 * channels * itemsize, y0, out_t)
+        return out_t.numpy()
     out = np.empty((h, w * channels), dtype=dtype.numpy_dtype)
     with open(path, "rb") as f:
         for r in range(h):
@@ -83,6 +104,15 @@ def write_block(
     total = height * row_elems * itemsize
     # Ensure the file exists at full size (rank-parallel writers each own
     # disjoint regions; os.O_CREAT + pwrite keeps them independent).
+    C = _native_io()
+    if C is not None:
+        import torch
+
+        blk_t = torch.from_numpy(np.ascontiguousarray(block))
+        C.write_block(
+            path, row_elems * itemsize, total, x0 * channels * itemsize, y0, blk_t
+        )
+        return
     fd = os.open(path, os.O_RDWR | os.O_CREAT, 0o644)
     try:
         if os.fstat(fd).st_size < total:
