@@ -61,6 +61,13 @@ void launch_conv(DType dt, int S, int k, const void* src, void* dst,
                  const float* wgt, const BlockGeom& g, const Region& r,
                  bool useMfma, hipStream_t stream);
 
+// Integer-dot u8 fast path (k=3, S=1, non-negative dyadic filter
+// w = n/2^shift, n in [0,255], sum(n) <= 2^shift): exact vs the fp32
+// contract. w9 = the nine integer taps row-major.
+void launch_conv_u8dot(const void* src, void* dst, const unsigned* w9,
+                       int shift, const BlockGeom& g, const Region& r,
+                       hipStream_t stream);
+
 // Fused double iteration: iteration A into an LDS mid image (quantized),
 // iteration B to global — rm is iteration A's region, r the output
 // region. Returns false if no fused template exists for (k, S).

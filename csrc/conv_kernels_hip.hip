@@ -368,6 +368,144 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
   }  // work loop
 }
 
+// ---- integer dot-product u8 kernel (headline fast path) -------------------
+//
+// For u8 grayscale with a NON-NEGATIVE DYADIC filter (w = n/2^k, n in
+// [0,255], sum(n) <= 2^k — the mandated Gaussian blur qualifies) the
+// whole iteration is exact integer arithmetic: s = sum(n_i * p_i) via
+// v_dot4_u32_u8 on v_perm-packed pixel bytes, then round-half-even
+// (s + half) >> k with an explicit even-tie fix — bit-identical to the
+// fp32-accumulate contract (the fp32 sums are exact for dyadic filters,
+// and RNE == integer half-even). Removes the u8->f32 staging conversion,
+// fp fma chain and fp quantize of the VALU kernel: LDS holds raw bytes
+// (6 KB vs 23 KB) and the hot loop is perm+dot4.
+__global__ __launch_bounds__(NTHREADS) void conv_u8dot_k(
+    const unsigned char* __restrict__ src, unsigned char* __restrict__ dst,
+    unsigned w0, unsigned w1, unsigned w2, int shift, int pitchE, int bufh,
+    int ex0, int ex1, int py0, int py1, int abase, int tilesX, int chainLen,
+    int nChains) {
+  constexpr int ROWS_IN = TILE_Y + 2;  // R = 1
+  constexpr int PITCHB = WIN_X + 16;   // 304 B: 16-B aligned rows
+  constexpr int QPR = WIN_X / 16;      // 18 16-B quarters per row
+  constexpr int ITEMS = ROWS_IN * QPR; // 180 <= 256: one item per thread
+
+  __shared__ unsigned char lds[2][ROWS_IN][PITCHB];
+
+  const int tid = threadIdx.x;
+  const unsigned wrow[3] = {w0, w1, w2};
+  const unsigned half = 1u << (shift - 1);
+  const unsigned mask = (1u << shift) - 1u;
+
+  const int nWork = tilesX * nChains;
+  for (int work = blockIdx.x; work < nWork; work += gridDim.x) {
+    const int bx = work % tilesX;
+    const int bc = work / tilesX;
+    const int etx0 = abase + TILE_X * bx;
+    const int gx0 = etx0 - 16;
+    const int tyBase = py0 + TILE_Y * chainLen * bc;
+    const int nt = min(chainLen, (py1 - tyBase + TILE_Y - 1) / TILE_Y);
+
+    const int srow = tid / QPR;
+    const int sq = tid % QPR;
+    const bool stager = tid < ITEMS;
+    const int ga = max(0, min(gx0 + 16 * sq, pitchE - 16));
+
+    auto issue = [&](uint4& pre, int tile) {
+      if (!stager) return;
+      const int gy = min(max(tyBase + tile * TILE_Y - 1 + srow, 0), bufh - 1);
+      pre = *reinterpret_cast<const uint4*>(src + (size_t)gy * pitchE + ga);
+    };
+    auto writeB = [&](const uint4& pre, int buf) {
+      if (!stager) return;
+      *reinterpret_cast<uint4*>(&lds[buf][srow][16 * sq]) = pre;
+    };
+
+    const int tx = tid & 63;
+    const int tyy = tid >> 6;
+    const int ex = etx0 + VEC * tx;
+
+    auto compute_store = [&](int t, int buf) {
+      const int ty0 = tyBase + t * TILE_Y;
+      // 12-byte window per input row: bytes [4tx+12, 4tx+24) of the
+      // staged row; out px p uses local bytes (3+p, 4+p, 5+p)
+      unsigned d[4][3];
+#pragma unroll
+      for (int rw = 0; rw < 4; ++rw) {
+        const unsigned* rp = reinterpret_cast<const unsigned*>(
+            &lds[buf][2 * tyy + rw][4 * tx + 12]);
+        d[rw][0] = rp[0];
+        d[rw][1] = rp[1];
+        d[rw][2] = rp[2];
+      }
+      // packed 4-byte pixel groups per (row, p): taps at local bytes
+      // 3+p..5+p; 4th byte is junk (its weight byte is 0)
+      unsigned pk[4][4];
+#pragma unroll
+      for (int rw = 0; rw < 4; ++rw) {
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+          if (p < 3) {
+            const unsigned sel =
+                (unsigned)(3 + p) | ((unsigned)(4 + p) << 8) |
+                ((unsigned)(5 + p) << 16);
+            // {S0:high, S1:low}: byte index 0-3 = S1, 4-7 = S0
+            pk[rw][p] = __builtin_amdgcn_perm(d[rw][1], d[rw][0], sel);
+          } else {
+            const unsigned sel = 2u | (3u << 8) | (4u << 16);
+            pk[rw][p] = __builtin_amdgcn_perm(d[rw][2], d[rw][1], sel);
+          }
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const int y = ty0 + 2 * tyy + j;
+        if (y >= py1) continue;
+        unsigned qb = 0;
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+          unsigned s = half;
+#pragma unroll
+          for (int rw = 0; rw < 3; ++rw)
+            s = __builtin_amdgcn_udot4(pk[j + rw][p], wrow[rw], s, false);
+          unsigned q = s >> shift;
+          // half-even: s already includes +half, so a tie is (s&mask)==0
+          q -= ((s & mask) == 0u) & q & 1u;
+          qb |= q << (8 * p);
+        }
+        unsigned char* out = dst + (size_t)y * pitchE + ex;
+        if (ex >= ex0 && ex + VEC <= ex1) {
+          *reinterpret_cast<unsigned*>(out) = qb;
+        } else {
+#pragma unroll
+          for (int p = 0; p < VEC; ++p)
+            if (ex + p >= ex0 && ex + p < ex1)
+              out[p] = (unsigned char)((qb >> (8 * p)) & 0xffu);
+        }
+      }
+    };
+
+    uint4 preA, preB;
+    issue(preA, 0);
+    writeB(preA, 0);
+    if (nt > 1) issue(preA, 1);
+    __syncthreads();
+
+    for (int tt = 0; tt < nt; tt += 2) {
+      if (tt + 2 < nt) issue(preB, tt + 2);
+      compute_store(tt, 0);
+      if (tt + 1 < nt) writeB(preA, 1);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      if (tt + 1 >= nt) break;
+      if (tt + 3 < nt) issue(preA, tt + 3);
+      compute_store(tt + 1, 1);
+      if (tt + 2 < nt) writeB(preB, 0);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+}
+
 // ---- fused double-iteration kernel ----------------------------------------
 //
 // Two convolution iterations per launch: iteration A is computed into an
@@ -991,7 +1129,33 @@ bool launch2_dtype(int S, int k, const void* src, void* dst, const float* wgt,
   return false;
 }
 
+void launch_u8dot_impl(const void* src, void* dst, const unsigned* w9,
+                       int shift, const BlockGeom& g, const Region& r,
+                       hipStream_t stream) {
+  const int abase = (r.ex0 / 16) * 16;
+  const int tilesX = (r.ex1 - abase + TILE_X - 1) / TILE_X;
+  const int tilesY = (r.py1 - r.py0 + TILE_Y - 1) / TILE_Y;
+  int chainLen = std::max(1, std::min(16, (tilesX * tilesY) / 4096));
+  if (chain_override() > 0) chainLen = chain_override();
+  const int chains = (tilesY + chainLen - 1) / chainLen;
+  const int grid = grid_for((const void*)&conv_u8dot_k, tilesX * chains);
+  const unsigned w0 = w9[0] | (w9[1] << 8) | (w9[2] << 16);
+  const unsigned w1 = w9[3] | (w9[4] << 8) | (w9[5] << 16);
+  const unsigned w2 = w9[6] | (w9[7] << 8) | (w9[8] << 16);
+  hipLaunchKernelGGL(conv_u8dot_k, dim3(grid), dim3(NTHREADS), 0, stream,
+                     (const unsigned char*)src, (unsigned char*)dst, w0, w1,
+                     w2, shift, g.pitchE, g.bufh, r.ex0, r.ex1, r.py0, r.py1,
+                     abase, tilesX, chainLen, chains);
+}
+
 }  // namespace
+
+void launch_conv_u8dot(const void* src, void* dst, const unsigned* w9,
+                       int shift, const BlockGeom& g, const Region& r,
+                       hipStream_t stream) {
+  if (r.py1 <= r.py0 || r.ex1 <= r.ex0) return;
+  launch_u8dot_impl(src, dst, w9, shift, g, r, stream);
+}
 
 void launch_conv(DType dt, int S, int k, const void* src, void* dst,
                  const float* wgt, const BlockGeom& g, const Region& r,

@@ -59,8 +59,12 @@ struct DeviceBlock {
 class ConvEngine {
  public:
   ConvEngine(int dtype, int k, const std::vector<float>& filter, int channels,
-             bool useMfma = false)
+             bool useMfma = false, std::vector<int> dotW = {})
       : dt_(static_cast<DType>(dtype)), k_(k), S_(channels), useMfma_(useMfma) {
+    if (dotW.size() == 10 && dt_ == DType::U8 && k == 3 && channels == 1) {
+      dotShift_ = dotW[0];
+      for (int i = 0; i < 9; ++i) dotW_[i] = (unsigned)dotW[1 + i];
+    }
     if ((int)filter.size() != k * k) throw std::runtime_error("filter size != k*k");
     HIP_CHECK(hipStreamCreateWithFlags(&compute_, hipStreamNonBlocking));
     HIP_CHECK(hipStreamCreateWithFlags(&comm_, hipStreamNonBlocking));
@@ -156,10 +160,7 @@ class ConvEngine {
         // border strips of region(0) launch after the exchange event.
         HIP_CHECK(hipEventRecord(evCompute_, compute_));
         HIP_CHECK(hipStreamWaitEvent(comm_, evCompute_, 0));
-        for (auto& b : blocks_) {
-          launch_conv(dt_, S_, k_, b.src, b.dst, (const float*)wgt_, b.g,
-                      interior_region(b.g), useMfma_, compute_);
-        }
+        for (auto& b : blocks_) launch_region(b, interior_region(b.g));
         if (comm_nccl_) {
           exchange_nccl(d);
         } else {
@@ -179,9 +180,7 @@ class ConvEngine {
               {in.py0, in.py1, full.ex0, in.ex0},
               {in.py0, in.py1, in.ex1, full.ex1},
           };
-          for (const Region& s : strips)
-            launch_conv(dt_, S_, k_, b.src, b.dst, (const float*)wgt_, b.g, s,
-                        useMfma_, compute_);
+          for (const Region& s : strips) launch_region(b, s);
         }
         for (auto& b : blocks_) std::swap(b.src, b.dst);
         swapped_ = !swapped_;
@@ -209,10 +208,7 @@ class ConvEngine {
           }
         }
         const int ext = d - (t + 1) * r;
-        for (auto& b : blocks_) {
-          launch_conv(dt_, S_, k_, b.src, b.dst, (const float*)wgt_, b.g,
-                      region_for(b.g, ext), useMfma_, compute_);
-        }
+        for (auto& b : blocks_) launch_region(b, region_for(b.g, ext));
         for (auto& b : blocks_) std::swap(b.src, b.dst);
         swapped_ = !swapped_;
         ++t;
@@ -281,6 +277,17 @@ class ConvEngine {
     rg.ex0 = (g.haloPx + r) * g.S;
     rg.ex1 = std::max((g.haloPx + g.wPx - r) * g.S, rg.ex0);
     return rg;
+  }
+
+  // single-iteration launch: integer-dot u8 fast path when eligible,
+  // else MFMA/VALU via launch_conv
+  void launch_region(DeviceBlock& b, const Region& rg) {
+    if (dotShift_ > 0) {
+      launch_conv_u8dot(b.src, b.dst, dotW_, dotShift_, b.g, rg, compute_);
+    } else {
+      launch_conv(dt_, S_, k_, b.src, b.dst, (const float*)wgt_, b.g, rg,
+                  useMfma_, compute_);
+    }
   }
 
   Region region_for(const BlockGeom& g, int ext) const {
@@ -409,6 +416,8 @@ class ConvEngine {
   bool swapped_ = false;
   bool useMfma_ = false;
   bool fusePairs_ = false;
+  int dotShift_ = 0;
+  unsigned dotW_[9] = {};
   bool phaseTiming_ = false;
   hipEvent_t evPhase_[4] = {};
   double exchangeMs_ = 0.0;

@@ -53,6 +53,31 @@ def mfma_eligible(spec: ProblemSpec) -> bool:
     return spec.k == 5
 
 
+def u8dot_weights(spec: ProblemSpec):
+    """Integer taps [shift, n0..n8] for the exact u8 dot-product path
+    (csrc conv_u8dot_k): k=3 grayscale with a non-negative dyadic filter
+    w = n/2^shift, n in [0,255], sum(n) <= 2^shift — the mandated
+    Gaussian blur qualifies. Empty list = ineligible.
+    PCONV_DISABLE_DOT=1 forces the fp32 VALU path for A/B."""
+    import os
+
+    if os.environ.get("PCONV_DISABLE_DOT"):
+        return []
+    if spec.dtype != DType.U8 or spec.k != 3 or spec.channels != 1:
+        return []
+    w = spec.filter.astype("float64").reshape(-1)
+    if (w < 0).any():
+        return []
+    for shift in range(1, 16):
+        n = w * (1 << shift)
+        if (n == n.round()).all():
+            n = n.round().astype("int64")
+            if n.max() <= 255 and n.sum() <= (1 << shift):
+                return [shift] + [int(v) for v in n]
+            return []
+    return []
+
+
 def load_extension():
     try:
         from . import _C  # noqa: F401
@@ -91,8 +116,14 @@ class HipEngine:
         self.blocks: List[_HipBlock] = []
         filt = [float(v) for v in spec.filter.flat]
         self.use_mfma = mfma_eligible(spec)
+        self.dot_weights = u8dot_weights(spec)
         self.engine = self._C.ConvEngine(
-            _DT_CODE[spec.dtype], spec.k, filt, spec.channels, self.use_mfma
+            _DT_CODE[spec.dtype],
+            spec.k,
+            filt,
+            spec.channels,
+            self.use_mfma,
+            self.dot_weights,
         )
         for bs in blocks:
             shape = (bs.buf_h, bs.buf_w_elems)
