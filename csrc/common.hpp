@@ -68,6 +68,11 @@ void launch_conv_u8dot(const void* src, void* dst, const unsigned* w9,
                        int shift, const BlockGeom& g, const Region& r,
                        hipStream_t stream);
 
+// Fused integer-dot double iteration (u8 mid image in LDS)
+void launch_conv_u8dot2(const void* src, void* dst, const unsigned* w9,
+                        int shift, const BlockGeom& g, const Region& rm,
+                        const Region& ro, hipStream_t stream);
+
 // Fused double iteration: iteration A into an LDS mid image (quantized),
 // iteration B to global — rm is iteration A's region, r the output
 // region. Returns false if no fused template exists for (k, S).

@@ -506,6 +506,171 @@ __global__ __launch_bounds__(NTHREADS) void conv_u8dot_k(
   }
 }
 
+// ---- fused integer-dot u8 double-iteration kernel -------------------------
+//
+// conv_u8dot_k x conv_tile2_k: two exact-integer iterations per launch
+// with the intermediate u8 image in LDS (6.7 KB total) — for the
+// headline filter class when the footprint exceeds the 256 MiB L3
+// (HBM-bound regime; L3-resident blocks keep the single-step kernel).
+__global__ __launch_bounds__(NTHREADS) void conv_u8dot2_k(
+    const unsigned char* __restrict__ src, unsigned char* __restrict__ dst,
+    unsigned w0, unsigned w1, unsigned w2, int shift, int pitchE, int bufh,
+    int ex0, int ex1, int py0, int py1, int m_ex0, int m_ex1, int m_py0,
+    int m_py1, int abase, int tilesX, int chainLen, int nChains) {
+  constexpr int ROWS_IN = TILE_Y + 4;   // staged input rows (2 iterations)
+  constexpr int ROWS_MID = TILE_Y + 2;  // iteration-A rows
+  constexpr int PITCHB = WIN_X + 16;    // 304 B rows
+  constexpr int QPR = WIN_X / 16;
+  constexpr int ITEMS = ROWS_IN * QPR;  // 216 <= 256
+
+  struct Smem {
+    unsigned char img[ROWS_IN][PITCHB];
+    unsigned char mid[ROWS_MID][PITCHB];
+  };
+  __shared__ Smem sm;
+
+  const int tid = threadIdx.x;
+  const unsigned wrow[3] = {w0, w1, w2};
+  const unsigned half = 1u << (shift - 1);
+  const unsigned mask = (1u << shift) - 1u;
+
+  // 3 consecutive window bytes (s, s+1, s+2 of {hi:lo}) + junk byte 3
+  auto pick3 = [](unsigned lo, unsigned hi, int st) {
+    const unsigned sel =
+        (unsigned)st | ((unsigned)(st + 1) << 8) | ((unsigned)(st + 2) << 16);
+    return __builtin_amdgcn_perm(hi, lo, sel);
+  };
+  // one 9-tap dot at local byte base 3+p of a 12-byte window (3 dwords
+  // per row), rounded half-even
+  auto dot9 = [&](const unsigned (&dr)[3][3], int p) {
+    unsigned s = half;
+#pragma unroll
+    for (int rw = 0; rw < 3; ++rw) {
+      unsigned pk = (p < 3) ? pick3(dr[rw][0], dr[rw][1], 3 + p)
+                            : pick3(dr[rw][1], dr[rw][2], 2);
+      s = __builtin_amdgcn_udot4(pk, wrow[rw], s, false);
+    }
+    unsigned q = s >> shift;
+    q -= ((s & mask) == 0u) & q & 1u;
+    return q;
+  };
+
+  const int nWork = tilesX * nChains;
+  for (int work = blockIdx.x; work < nWork; work += gridDim.x) {
+    const int bx = work % tilesX;
+    const int bc = work / tilesX;
+    const int etx0 = abase + TILE_X * bx;
+    const int gx0 = etx0 - 16;
+    const int tyBase = py0 + TILE_Y * chainLen * bc;
+    const int nt = min(chainLen, (py1 - tyBase + TILE_Y - 1) / TILE_Y);
+
+    const int srow = tid / QPR;
+    const int sq = tid % QPR;
+    const bool stager = tid < ITEMS;
+    const int ga = max(0, min(gx0 + 16 * sq, pitchE - 16));
+
+    uint4 pre;
+    auto issue = [&](int tile) {
+      if (!stager) return;
+      const int gy = min(max(tyBase + tile * TILE_Y - 2 + srow, 0), bufh - 1);
+      pre = *reinterpret_cast<const uint4*>(src + (size_t)gy * pitchE + ga);
+    };
+    auto writeB = [&]() {
+      if (!stager) return;
+      *reinterpret_cast<uint4*>(&sm.img[srow][16 * sq]) = pre;
+    };
+
+    const int tx = tid & 63;
+    const int tyy = tid >> 6;
+    const int ex = etx0 + VEC * tx;
+
+    issue(0);
+    writeB();
+    if (nt > 1) issue(1);
+    __syncthreads();
+
+    for (int t = 0; t < nt; ++t) {
+      const int ty0 = tyBase + t * TILE_Y;
+      const bool pf = t + 1 < nt;
+
+      // ---- iteration A: dot(img) -> quantized u8 mid ----
+      // mid chunk c covers elements etx0-4+4c (window bytes 12+4c..),
+      // rows r1 absolute ty0 - 1 + r1; input rows r1..r1+2 of img.
+      auto iterA = [&](int r1, int c) {
+        unsigned dr[3][3];
+#pragma unroll
+        for (int rw = 0; rw < 3; ++rw) {
+          const unsigned* rp =
+              reinterpret_cast<const unsigned*>(&sm.img[r1 + rw][4 * c + 8]);
+          dr[rw][0] = rp[0];
+          dr[rw][1] = rp[1];
+          dr[rw][2] = rp[2];
+        }
+        const int y1 = ty0 - 1 + r1;
+        const bool rowIn = y1 >= m_py0 && y1 < m_py1;
+        const int xe = etx0 - 4 + 4 * c;
+        unsigned qb = 0;
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+          unsigned q = dot9(dr, p);
+          const bool in = rowIn && xe + p >= m_ex0 && xe + p < m_ex1;
+          qb |= (in ? q : 0u) << (8 * p);
+        }
+        *reinterpret_cast<unsigned*>(&sm.mid[r1][12 + 4 * c]) = qb;
+      };
+#pragma unroll
+      for (int cpass = 0; cpass < 2; ++cpass) {
+        const int c = tx + 64 * cpass;
+        if (cpass == 1 && c >= 66) break;
+        iterA(2 * tyy, c);
+        iterA(2 * tyy + 1, c);
+        if (tyy < 2) iterA(8 + tyy, c);  // leftover rows, spread over waves
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      if (pf) writeB();           // land next tile's input (img is free)
+      if (t + 2 < nt) issue(t + 2);  // prefetch lands during B + next A
+
+      // ---- iteration B: dot(mid) -> global ----
+      {
+        unsigned d[4][3];
+#pragma unroll
+        for (int rw = 0; rw < 4; ++rw) {
+          const unsigned* rp = reinterpret_cast<const unsigned*>(
+              &sm.mid[2 * tyy + rw][4 * tx + 12]);
+          d[rw][0] = rp[0];
+          d[rw][1] = rp[1];
+          d[rw][2] = rp[2];
+        }
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+          const int y = ty0 + 2 * tyy + j;
+          if (y >= py1) continue;
+          unsigned dr[3][3];
+#pragma unroll
+          for (int rw = 0; rw < 3; ++rw)
+#pragma unroll
+            for (int i = 0; i < 3; ++i) dr[rw][i] = d[j + rw][i];
+          unsigned qb = 0;
+#pragma unroll
+          for (int p = 0; p < 4; ++p) qb |= dot9(dr, p) << (8 * p);
+          unsigned char* out = dst + (size_t)y * pitchE + ex;
+          if (ex >= ex0 && ex + VEC <= ex1) {
+            *reinterpret_cast<unsigned*>(out) = qb;
+          } else {
+#pragma unroll
+            for (int p = 0; p < VEC; ++p)
+              if (ex + p >= ex0 && ex + p < ex1)
+                out[p] = (unsigned char)((qb >> (8 * p)) & 0xffu);
+          }
+        }
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+}
+
 // ---- fused double-iteration kernel ----------------------------------------
 //
 // Two convolution iterations per launch: iteration A is computed into an
@@ -1148,7 +1313,34 @@ void launch_u8dot_impl(const void* src, void* dst, const unsigned* w9,
                      abase, tilesX, chainLen, chains);
 }
 
+void launch_u8dot2_impl(const void* src, void* dst, const unsigned* w9,
+                        int shift, const BlockGeom& g, const Region& rm,
+                        const Region& ro, hipStream_t stream) {
+  const int abase = (ro.ex0 / 16) * 16;
+  const int tilesX = (ro.ex1 - abase + TILE_X - 1) / TILE_X;
+  const int tilesY = (ro.py1 - ro.py0 + TILE_Y - 1) / TILE_Y;
+  int chainLen = std::max(1, std::min(16, (tilesX * tilesY) / 4096));
+  if (chain_override() > 0) chainLen = chain_override();
+  const int chains = (tilesY + chainLen - 1) / chainLen;
+  const int grid = grid_for((const void*)&conv_u8dot2_k, tilesX * chains);
+  const unsigned w0 = w9[0] | (w9[1] << 8) | (w9[2] << 16);
+  const unsigned w1 = w9[3] | (w9[4] << 8) | (w9[5] << 16);
+  const unsigned w2 = w9[6] | (w9[7] << 8) | (w9[8] << 16);
+  hipLaunchKernelGGL(conv_u8dot2_k, dim3(grid), dim3(NTHREADS), 0, stream,
+                     (const unsigned char*)src, (unsigned char*)dst, w0, w1,
+                     w2, shift, g.pitchE, g.bufh, ro.ex0, ro.ex1, ro.py0,
+                     ro.py1, rm.ex0, rm.ex1, rm.py0, rm.py1, abase, tilesX,
+                     chainLen, chains);
+}
+
 }  // namespace
+
+void launch_conv_u8dot2(const void* src, void* dst, const unsigned* w9,
+                        int shift, const BlockGeom& g, const Region& rm,
+                        const Region& ro, hipStream_t stream) {
+  if (ro.py1 <= ro.py0 || ro.ex1 <= ro.ex0) return;
+  launch_u8dot2_impl(src, dst, w9, shift, g, rm, ro, stream);
+}
 
 void launch_conv_u8dot(const void* src, void* dst, const unsigned* w9,
                        int shift, const BlockGeom& g, const Region& r,

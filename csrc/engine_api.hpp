@@ -119,6 +119,13 @@ class ConvEngine {
       HIP_CHECK(hipMalloc(&b.recvE, strip));
     }
     blocks_.push_back(b);
+    // fused integer-dot pairs (conv_u8dot2_k) halve HBM traffic but the
+    // redundant iteration-A apron costs more than the bandwidth saved —
+    // measured 1256 vs 1658 Gpx-it/s at 16384² (the dot path is
+    // issue-bound, not BW-bound). Off by default; PCONV_DOT_FUSE=1
+    // selects it for A/B.
+    const char* de = getenv("PCONV_DOT_FUSE");
+    dotFuse_ = de ? atoi(de) != 0 : false;
     return (int)blocks_.size() - 1;
   }
 
@@ -191,14 +198,21 @@ class ConvEngine {
       if (phaseTiming_) HIP_CHECK(hipEventRecord(evPhase_[2], compute_));
       const bool canFuse =
           fusePairs_ && !useMfma_ && r >= 1 && r <= 2 && (S_ == 1 || S_ == 3);
+      const bool canDotFuse = dotFuse_ && dotShift_ > 0;
       for (int t = t0; t < T;) {
-        if (canFuse && t + 1 < T) {
+        if ((canFuse || canDotFuse) && t + 1 < T) {
           // fused pair: iteration t into LDS mid, iteration t+1 to global
           bool ok = true;
           for (auto& b : blocks_) {
-            ok &= launch_conv2(dt_, S_, k_, b.src, b.dst, (const float*)wgt_,
-                               b.g, region_for(b.g, d - (t + 1) * r),
-                               region_for(b.g, d - (t + 2) * r), compute_);
+            const Region rm = region_for(b.g, d - (t + 1) * r);
+            const Region ro = region_for(b.g, d - (t + 2) * r);
+            if (canDotFuse) {
+              launch_conv_u8dot2(b.src, b.dst, dotW_, dotShift_, b.g, rm, ro,
+                                 compute_);
+            } else {
+              ok &= launch_conv2(dt_, S_, k_, b.src, b.dst,
+                                 (const float*)wgt_, b.g, rm, ro, compute_);
+            }
           }
           if (ok) {
             for (auto& b : blocks_) std::swap(b.src, b.dst);
@@ -417,6 +431,7 @@ class ConvEngine {
   bool useMfma_ = false;
   bool fusePairs_ = false;
   int dotShift_ = 0;
+  bool dotFuse_ = false;
   unsigned dotW_[9] = {};
   bool phaseTiming_ = false;
   hipEvent_t evPhase_[4] = {};

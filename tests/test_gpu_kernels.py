@@ -425,3 +425,20 @@ def test_fused_f16_tie_rounding_consistency():
             del os.environ["PCONV_FUSE"]
     np.testing.assert_array_equal(outs["0"], outs["1"])
     np.testing.assert_array_equal(outs["0"], ref)
+
+
+@pytest.mark.parametrize("p", [1, 4])
+def test_fused_dot_byte_exact(p):
+    """Fused integer-dot pair path (PCONV_DOT_FUSE=1) bitwise vs oracle,
+    odd iteration count, multi-block exchange."""
+    import os
+
+    spec = ProblemSpec(96, 80, 1, DType.U8, GAUSSIAN_BLUR_3, 21)
+    img = synthetic_image(80, 96, 1, DType.U8)
+    os.environ["PCONV_DOT_FUSE"] = "1"
+    try:
+        out = _engine_mod().run_local_blocks_gpu(spec, img, p)
+    finally:
+        del os.environ["PCONV_DOT_FUSE"]
+    ref = run_reference(spec, img)
+    np.testing.assert_array_equal(out, ref)
