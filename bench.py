@@ -177,6 +177,9 @@ def main() -> int:
                 "seq_len": None,
                 "parallelism": "spatial2d-{}x{}".format(*blocks[0].grid),
                 "halo_chunk_iters": eng.chunk_iters,
+                "kernel_path": (
+                    eng.kernel_path() if hasattr(eng, "kernel_path") else "cpu"
+                ),
                 "preset": args.preset,
             },
         }

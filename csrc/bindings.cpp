@@ -95,6 +95,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("diff_max", &pconv::ConvEngine::diff_max)
       .def("swapped", &pconv::ConvEngine::swapped)
       .def("set_phase_timing", &pconv::ConvEngine::set_phase_timing)
+      .def("kernel_path", &pconv::ConvEngine::kernel_path)
       .def("phase_times", &pconv::ConvEngine::phase_times)
       .def("synchronize", &pconv::ConvEngine::synchronize);
 

@@ -245,6 +245,16 @@ class ConvEngine {
   }
 
   void set_phase_timing(bool on) { phaseTiming_ = on; }
+
+  // which compute path run() will take (observability: lands in the
+  // bench JSON so a measurement names the kernel that produced it)
+  std::string kernel_path() const {
+    if (dotShift_ > 0) return dotFuse_ ? "u8dot-fused" : "u8dot";
+    if (useMfma_) return "mfma-f16";
+    if (fusePairs_ && k_ / 2 >= 1 && k_ / 2 <= 2 && (S_ == 1 || S_ == 3))
+      return "valu-fused2";
+    return "valu";
+  }
   // (exchange_ms, compute_ms, chunks) accumulated since construction
   std::vector<double> phase_times() const {
     return {exchangeMs_, computeMs_, (double)chunks_};

@@ -211,6 +211,11 @@ class HipEngine:
         §5.1); adds one device sync per chunk — keep off in timed bench."""
         self.engine.set_phase_timing(on)
 
+    def kernel_path(self) -> str:
+        """Which compute path run() takes (u8dot / mfma-f16 / valu-fused2
+        / valu) — recorded in bench output."""
+        return str(self.engine.kernel_path())
+
     def phase_times(self) -> dict:
         ex, comp, chunks = self.engine.phase_times()
         return {"exchange_ms": ex, "compute_ms": comp, "chunks": int(chunks)}
