@@ -28,6 +28,20 @@
 //   * Tiles snap to 16-element x-origins so every staging load stays
 //     16-B-aligned regardless of the (halo-deepened) region offset.
 
+// Contents (one compile unit so the dtype traits and tile geometry are
+// shared; every kernel is selected per problem by the engine — see
+// ConvEngine::kernel_path and hip_engine.mfma_eligible/u8dot_weights):
+//   U8T/F16T/F32T        dtype traits (loads, stores, quantization)
+//   conv_u8dot_k         exact integer v_dot4 path (dyadic u8 3x3) [default: headline]
+//   conv_u8dot2_k        fused-pair variant of the above [PCONV_DOT_FUSE=1 only]
+//   conv_tile2_k         fused double iteration, fp mid image in LDS [default: fp16/fp32]
+//   conv_tile_k          fp32-accumulate VALU kernel [default: everything else]
+//   conv_mfma_k          fp16-input MFMA, banded-Toeplitz filter operand [default: k=5]
+//   mfma_probe_k         fragment-layout verification probe
+//   conv_generic_k       any odd k / channel stride fallback
+//   diff_max_k           convergence reduction
+//   launch_* dispatch    region tiling, chain scheduling, occupancy-sized grids
+
 #include "common.hpp"
 
 #include <hip/hip_fp16.h>
