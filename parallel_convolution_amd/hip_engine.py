@@ -202,7 +202,10 @@ class HipEngine:
     def diff_max(self) -> float:
         """max |last - previous| over owned regions of local blocks (the
         convergence test of SURVEY.md §2.1 R10; allreduce is the caller's
-        job in distributed mode)."""
+        job in distributed mode). When the last step ran a fused
+        double-iteration kernel, "previous" is two iterations back — a
+        conservative (larger) delta, so convergence may be detected one
+        check later but never falsely early."""
         torch.cuda.synchronize()
         return float(self.engine.diff_max())
 
