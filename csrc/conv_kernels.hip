@@ -392,15 +392,19 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
 // and RNE == integer half-even). Removes the u8->f32 staging conversion,
 // fp fma chain and fp quantize of the VALU kernel: LDS holds raw bytes
 // (6 KB vs 23 KB) and the hot loop is perm+dot4.
+template <int CP>  // compute passes: tile covers 256*CP output elements
 __global__ __launch_bounds__(NTHREADS) void conv_u8dot_k(
     const unsigned char* __restrict__ src, unsigned char* __restrict__ dst,
     unsigned w0, unsigned w1, unsigned w2, int shift, int pitchE, int bufh,
     int ex0, int ex1, int py0, int py1, int abase, int tilesX, int chainLen,
     int nChains) {
-  constexpr int ROWS_IN = TILE_Y + 2;  // R = 1
-  constexpr int PITCHB = WIN_X + 16;   // 304 B: 16-B aligned rows
-  constexpr int QPR = WIN_X / 16;      // 18 16-B quarters per row
-  constexpr int ITEMS = ROWS_IN * QPR; // 180 <= 256: one item per thread
+  constexpr int ROWS_IN = TILE_Y + 2;        // R = 1
+  constexpr int TXE = 256 * CP;              // tile elements
+  constexpr int WINB = TXE + 32;             // staged window bytes/row
+  constexpr int PITCHB = WINB + 16;          // 16-B aligned rows
+  constexpr int QPR = WINB / 16;
+  constexpr int ITEMS = ROWS_IN * QPR;
+  constexpr int NITER = (ITEMS + NTHREADS - 1) / NTHREADS;
 
   __shared__ unsigned char lds[2][ROWS_IN][PITCHB];
 
@@ -413,39 +417,56 @@ __global__ __launch_bounds__(NTHREADS) void conv_u8dot_k(
   for (int work = blockIdx.x; work < nWork; work += gridDim.x) {
     const int bx = work % tilesX;
     const int bc = work / tilesX;
-    const int etx0 = abase + TILE_X * bx;
+    const int etx0 = abase + TXE * bx;
     const int gx0 = etx0 - 16;
     const int tyBase = py0 + TILE_Y * chainLen * bc;
     const int nt = min(chainLen, (py1 - tyBase + TILE_Y - 1) / TILE_Y);
 
-    const int srow = tid / QPR;
-    const int sq = tid % QPR;
-    const bool stager = tid < ITEMS;
-    const int ga = max(0, min(gx0 + 16 * sq, pitchE - 16));
+    int srow[NITER], sq[NITER], ga[NITER];
+#pragma unroll
+    for (int k = 0; k < NITER; ++k) {
+      const int i = tid + k * NTHREADS;
+      srow[k] = i / QPR;
+      sq[k] = i % QPR;
+      ga[k] = max(0, min(gx0 + 16 * sq[k], pitchE - 16));
+    }
 
-    auto issue = [&](uint4& pre, int tile) {
-      if (!stager) return;
-      const int gy = min(max(tyBase + tile * TILE_Y - 1 + srow, 0), bufh - 1);
-      pre = *reinterpret_cast<const uint4*>(src + (size_t)gy * pitchE + ga);
+    auto issue = [&](uint4 (&pre)[NITER], int tile) {
+#pragma unroll
+      for (int k = 0; k < NITER; ++k) {
+        if (k == NITER - 1 && ITEMS % NTHREADS != 0 &&
+            tid >= ITEMS - k * NTHREADS)
+          break;
+        const int gy =
+            min(max(tyBase + tile * TILE_Y - 1 + srow[k], 0), bufh - 1);
+        pre[k] = *reinterpret_cast<const uint4*>(src + (size_t)gy * pitchE +
+                                                 ga[k]);
+      }
     };
-    auto writeB = [&](const uint4& pre, int buf) {
-      if (!stager) return;
-      *reinterpret_cast<uint4*>(&lds[buf][srow][16 * sq]) = pre;
+    auto writeB = [&](const uint4 (&pre)[NITER], int buf) {
+#pragma unroll
+      for (int k = 0; k < NITER; ++k) {
+        if (k == NITER - 1 && ITEMS % NTHREADS != 0 &&
+            tid >= ITEMS - k * NTHREADS)
+          break;
+        *reinterpret_cast<uint4*>(&lds[buf][srow[k]][16 * sq[k]]) = pre[k];
+      }
     };
 
     const int tx = tid & 63;
     const int tyy = tid >> 6;
-    const int ex = etx0 + VEC * tx;
 
-    auto compute_store = [&](int t, int buf) {
+    auto compute_pass = [&](int t, int buf, int cp) {
       const int ty0 = tyBase + t * TILE_Y;
-      // 12-byte window per input row: bytes [4tx+12, 4tx+24) of the
+      const int txc = tx + 64 * cp;
+      const int ex = etx0 + VEC * txc;
+      // 12-byte window per input row: bytes [4txc+12, 4txc+24) of the
       // staged row; out px p uses local bytes (3+p, 4+p, 5+p)
       unsigned d[4][3];
 #pragma unroll
       for (int rw = 0; rw < 4; ++rw) {
         const unsigned* rp = reinterpret_cast<const unsigned*>(
-            &lds[buf][2 * tyy + rw][4 * tx + 12]);
+            &lds[buf][2 * tyy + rw][4 * txc + 12]);
         d[rw][0] = rp[0];
         d[rw][1] = rp[1];
         d[rw][2] = rp[2];
@@ -497,7 +518,12 @@ __global__ __launch_bounds__(NTHREADS) void conv_u8dot_k(
       }
     };
 
-    uint4 preA, preB;
+    auto compute_store = [&](int t, int buf) {
+#pragma unroll
+      for (int cp = 0; cp < CP; ++cp) compute_pass(t, buf, cp);
+    };
+
+    uint4 preA[NITER], preB[NITER];
     issue(preA, 0);
     writeB(preA, 0);
     if (nt > 1) issue(preA, 1);
@@ -1307,23 +1333,44 @@ bool launch2_dtype(int S, int k, const void* src, void* dst, const float* wgt,
   return false;
 }
 
-void launch_u8dot_impl(const void* src, void* dst, const unsigned* w9,
-                       int shift, const BlockGeom& g, const Region& r,
-                       hipStream_t stream) {
+template <int CP>
+void launch_u8dot_cp(const void* src, void* dst, const unsigned* w9,
+                     int shift, const BlockGeom& g, const Region& r,
+                     hipStream_t stream) {
+  const int TXE = 256 * CP;
   const int abase = (r.ex0 / 16) * 16;
-  const int tilesX = (r.ex1 - abase + TILE_X - 1) / TILE_X;
+  const int tilesX = (r.ex1 - abase + TXE - 1) / TXE;
   const int tilesY = (r.py1 - r.py0 + TILE_Y - 1) / TILE_Y;
   int chainLen = std::max(1, std::min(16, (tilesX * tilesY) / 4096));
   if (chain_override() > 0) chainLen = chain_override();
   const int chains = (tilesY + chainLen - 1) / chainLen;
-  const int grid = grid_for((const void*)&conv_u8dot_k, tilesX * chains);
+  const int grid = grid_for((const void*)&conv_u8dot_k<CP>, tilesX * chains);
   const unsigned w0 = w9[0] | (w9[1] << 8) | (w9[2] << 16);
   const unsigned w1 = w9[3] | (w9[4] << 8) | (w9[5] << 16);
   const unsigned w2 = w9[6] | (w9[7] << 8) | (w9[8] << 16);
-  hipLaunchKernelGGL(conv_u8dot_k, dim3(grid), dim3(NTHREADS), 0, stream,
-                     (const unsigned char*)src, (unsigned char*)dst, w0, w1,
-                     w2, shift, g.pitchE, g.bufh, r.ex0, r.ex1, r.py0, r.py1,
-                     abase, tilesX, chainLen, chains);
+  hipLaunchKernelGGL((conv_u8dot_k<CP>), dim3(grid), dim3(NTHREADS), 0,
+                     stream, (const unsigned char*)src, (unsigned char*)dst,
+                     w0, w1, w2, shift, g.pitchE, g.bufh, r.ex0, r.ex1, r.py0,
+                     r.py1, abase, tilesX, chainLen, chains);
+}
+
+void launch_u8dot_impl(const void* src, void* dst, const unsigned* w9,
+                       int shift, const BlockGeom& g, const Region& r,
+                       hipStream_t stream) {
+  // 512-element tiles (CP=2) halve the barrier frequency but measured
+  // 31 % SLOWER at 16384² (1146 vs 1665 — the wider tile's staging
+  // register pressure and longer serial tail cost more than the barrier
+  // saved). CP=1 is the default; PCONV_DOT_CP=2 keeps the variant
+  // measurable.
+  static int ov = [] {
+    const char* e = getenv("PCONV_DOT_CP");
+    return e ? atoi(e) : 0;
+  }();
+  int cp = ov ? ov : 1;
+  if (cp >= 2)
+    launch_u8dot_cp<2>(src, dst, w9, shift, g, r, stream);
+  else
+    launch_u8dot_cp<1>(src, dst, w9, shift, g, r, stream);
 }
 
 void launch_u8dot2_impl(const void* src, void* dst, const unsigned* w9,
