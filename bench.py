@@ -45,6 +45,45 @@ PRESETS = {
 }
 
 
+def verify_distributed(rank: int, world: int) -> None:
+    from parallel_convolution_amd.hip_engine import HipEngine
+    from parallel_convolution_amd.ops.reference import run_reference
+    from parallel_convolution_amd.parallel.decomp import halo_for_radius
+
+    vspec = ProblemSpec(512, 512, 1, DType.U8, GAUSSIAN_BLUR_3, 6)
+    img = synthetic_image(512, 512, 1, DType.U8, seed=99)
+    vblocks = make_blocks(512, 512, world, 1, halo=halo_for_radius(1))
+    mine = vblocks[rank]
+    veng = HipEngine(vspec, [mine], world_blocks=vblocks)
+    veng.init_distributed(rank, world)
+    veng.load_block(0, img[mine.y0 : mine.y0 + mine.h, mine.x0 : mine.x0 + mine.w])
+    veng.run()
+    out = torch.from_numpy(veng.store_block(0))
+    # gather blocks to rank 0 over the gloo control group
+    if rank == 0:
+        full = np.zeros((512, 512), dtype=np.uint8)
+        full[mine.y0 : mine.y0 + mine.h, mine.x0 : mine.x0 + mine.w] = out.numpy()
+        for r in range(1, world):
+            b = vblocks[r]
+            t = torch.zeros((b.h, b.w), dtype=torch.uint8)
+            dist.recv(t, src=r)
+            full[b.y0 : b.y0 + b.h, b.x0 : b.x0 + b.w] = t.numpy()
+        ref = run_reference(vspec, img)
+        ok = bool(np.array_equal(full, ref))
+        bad = int((full != ref).sum())
+        print(
+            f"[verify] distributed 512x512 x6 over RCCL world={world}: "
+            f"{'PASS (bit-exact vs oracle)' if ok else f'FAIL ({bad} mismatches)'}",
+            file=sys.stderr,
+            flush=True,
+        )
+    else:
+        dist.send(out, dst=0)
+    dist.barrier()
+    del veng
+    torch.cuda.empty_cache()
+
+
 def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -55,6 +94,8 @@ def main() -> int:
     ap.add_argument("--json-out", default=None)
     ap.add_argument("--backend", choices=["hip", "cpu"], default="hip",
                     help="cpu = gloo protocol path (distributed-flow test only)")
+    ap.add_argument("--no-verify", action="store_true",
+                    help="skip the small-scale distributed correctness check")
     args = ap.parse_args()
 
     world = bootstrap.env_world()
@@ -120,6 +161,17 @@ def main() -> int:
                 cpu_eng.run(n)
 
         eng = _CpuShim()
+
+    # Small-scale correctness verify on the SAME world (512² decomposed,
+    # 6 iters, RCCL exchange) before the timed run: the oracle comparison
+    # lands in the bench log so a scaling number is never quoted off a
+    # silently-broken exchange. Costs ~1 s; --no-verify skips.
+    if world > 1 and args.backend == "hip" and not args.no_verify:
+        try:
+            verify_distributed(rank, world)
+        except Exception as e:  # the check must never break the bench
+            print(f"[verify] skipped ({type(e).__name__}: {e})",
+                  file=sys.stderr, flush=True)
 
     # synthetic random-pixel block data (BASELINE north-star), per-rank seed
     for i, bs in enumerate(my_blocks):
