@@ -1197,6 +1197,16 @@ inline int chain_override() {
   return v;
 }
 
+// chain length: amortize the per-chain staging prologue while keeping
+// enough work items to fill an occupancy-sized grid. Narrow regions
+// (few x-tiles — the 2x4-decomposed per-rank shape) prefer longer
+// chains: measured +5 % at 4096x8192 with 8 vs the square-tuned 4.
+inline int chain_for(int tilesX, int tilesY) {
+  if (chain_override() > 0) return chain_override();
+  const int denom = tilesX <= 16 ? 2048 : 4096;
+  return std::max(1, std::min(16, (tilesX * tilesY) / denom));
+}
+
 // occupancy-sized grid: one pass of resident workgroups grid-strides the
 // chain work items, so an awkward work count costs a remainder of TILES,
 // not a whole extra wave of chains
@@ -1231,10 +1241,7 @@ void launch_tile(const void* src, void* dst, const float* wgt,
   const int abase = (r.ex0 / 16) * 16;
   const int tilesX = (r.ex1 - abase + TILE_X - 1) / TILE_X;
   const int tilesY = (r.py1 - r.py0 + TILE_Y - 1) / TILE_Y;
-  // chain length: keep >= ~4096 workgroups in flight (256 CUs x 8 blocks
-  // with headroom); longer chains amortize the staging prologue.
-  int chainLen = std::max(1, std::min(16, (tilesX * tilesY) / 4096));
-  if (chain_override() > 0) chainLen = chain_override();
+  const int chainLen = chain_for(tilesX, tilesY);
   const int chains = (tilesY + chainLen - 1) / chainLen;
   const int grid =
       grid_for((const void*)&conv_tile_k<TT, R, S>, tilesX * chains);
@@ -1255,8 +1262,7 @@ void launch_mfma(const void* src, void* dst, const float* wgt,
   const int abase = (r.ex0 / 16) * 16;
   const int tilesX = (r.ex1 - abase + M_TILE_X - 1) / M_TILE_X;
   const int tilesY = (r.py1 - r.py0 + M_TILE_Y - 1) / M_TILE_Y;
-  int chainLen = std::max(1, std::min(16, (tilesX * tilesY) / 4096));
-  if (chain_override() > 0) chainLen = chain_override();
+  const int chainLen = chain_for(tilesX, tilesY);
   const int chains = (tilesY + chainLen - 1) / chainLen;
   const int grid =
       grid_for((const void*)&conv_mfma_k<TT, R, S>, tilesX * chains, M_NT);
@@ -1309,8 +1315,7 @@ void launch_tile2(const void* src, void* dst, const float* wgt,
   const int abase = (ro.ex0 / 16) * 16;
   const int tilesX = (ro.ex1 - abase + TILE_X - 1) / TILE_X;
   const int tilesY = (ro.py1 - ro.py0 + TILE_Y - 1) / TILE_Y;
-  int chainLen = std::max(1, std::min(16, (tilesX * tilesY) / 4096));
-  if (chain_override() > 0) chainLen = chain_override();
+  const int chainLen = chain_for(tilesX, tilesY);
   const int chains = (tilesY + chainLen - 1) / chainLen;
   const int grid =
       grid_for((const void*)&conv_tile2_k<TT, R, S>, tilesX * chains);
@@ -1341,8 +1346,7 @@ void launch_u8dot_cp(const void* src, void* dst, const unsigned* w9,
   const int abase = (r.ex0 / 16) * 16;
   const int tilesX = (r.ex1 - abase + TXE - 1) / TXE;
   const int tilesY = (r.py1 - r.py0 + TILE_Y - 1) / TILE_Y;
-  int chainLen = std::max(1, std::min(16, (tilesX * tilesY) / 4096));
-  if (chain_override() > 0) chainLen = chain_override();
+  const int chainLen = chain_for(tilesX, tilesY);
   const int chains = (tilesY + chainLen - 1) / chainLen;
   const int grid = grid_for((const void*)&conv_u8dot_k<CP>, tilesX * chains);
   const unsigned w0 = w9[0] | (w9[1] << 8) | (w9[2] << 16);
@@ -1379,8 +1383,7 @@ void launch_u8dot2_impl(const void* src, void* dst, const unsigned* w9,
   const int abase = (ro.ex0 / 16) * 16;
   const int tilesX = (ro.ex1 - abase + TILE_X - 1) / TILE_X;
   const int tilesY = (ro.py1 - ro.py0 + TILE_Y - 1) / TILE_Y;
-  int chainLen = std::max(1, std::min(16, (tilesX * tilesY) / 4096));
-  if (chain_override() > 0) chainLen = chain_override();
+  const int chainLen = chain_for(tilesX, tilesY);
   const int chains = (tilesY + chainLen - 1) / chainLen;
   const int grid = grid_for((const void*)&conv_u8dot2_k, tilesX * chains);
   const unsigned w0 = w9[0] | (w9[1] << 8) | (w9[2] << 16);
