@@ -1,5 +1,9 @@
 """Direct conv-kernel sweep across image sizes (cache-residency probe).
 
+Note: drives `_C.conv_region` (the VALU / MFMA single-kernel entry), not
+the engine's per-problem path selection — the engine may pick the
+integer-dot or fused kernels instead (see hip_engine.kernel_path).
+
 Sizes chosen so src+dst land in L2 (~32 MB), L3 (~256 MB), or HBM:
 2048²=8 MB, 4096²=33 MB, 8192²=134 MB, 16384²=537 MB (u8, two buffers).
 """
