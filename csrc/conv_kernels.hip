@@ -398,15 +398,20 @@ __global__ __launch_bounds__(NTHREADS) void conv_u8dot_k(
     unsigned w0, unsigned w1, unsigned w2, int shift, int pitchE, int bufh,
     int ex0, int ex1, int py0, int py1, int abase, int tilesX, int chainLen,
     int nChains) {
-  constexpr int ROWS_IN = TILE_Y + 2;        // R = 1
+  constexpr int ROWS_IN = TILE_Y + 2;        // rows a tile reads (R = 1)
+  constexpr int RING = 32;  // >= ROWS_IN + 2*TILE_Y in flight; pow2 slot
   constexpr int TXE = 256 * CP;              // tile elements
   constexpr int WINB = TXE + 32;             // staged window bytes/row
   constexpr int PITCHB = WINB + 16;          // 16-B aligned rows
   constexpr int QPR = WINB / 16;
-  constexpr int ITEMS = ROWS_IN * QPR;
-  constexpr int NITER = (ITEMS + NTHREADS - 1) / NTHREADS;
+  // steady state stages only TILE_Y new rows per tile (consecutive
+  // chain tiles share 2 rows): HBM reads drop from 1.25x to 1.0x in y
+  constexpr int SITEMS = TILE_Y * QPR;
+  constexpr int PITEMS = ROWS_IN * QPR;      // prologue stages all rows
+  constexpr int SNITER = (SITEMS + NTHREADS - 1) / NTHREADS;
+  constexpr int PNITER = (PITEMS + NTHREADS - 1) / NTHREADS;
 
-  __shared__ unsigned char lds[2][ROWS_IN][PITCHB];
+  __shared__ unsigned char lds[RING][PITCHB];
 
   const int tid = threadIdx.x;
   const unsigned wrow[3] = {w0, w1, w2};
@@ -422,57 +427,64 @@ __global__ __launch_bounds__(NTHREADS) void conv_u8dot_k(
     const int tyBase = py0 + TILE_Y * chainLen * bc;
     const int nt = min(chainLen, (py1 - tyBase + TILE_Y - 1) / TILE_Y);
 
-    int srow[NITER], sq[NITER], ga[NITER];
+    // ring slot of a chain-relative input row rr (rr = buffer row -
+    // (tyBase - 1)); rows enter the ring in order. The in-flight span
+    // (current tile's 10 rows + two prefetched tiles' 16) is < 32, so a
+    // power-of-2 ring makes the slot a single AND.
+    auto slot = [&](int rr) { return rr & (RING - 1); };
+
+    int srow[SNITER], sq[SNITER], ga[SNITER];
 #pragma unroll
-    for (int k = 0; k < NITER; ++k) {
+    for (int k = 0; k < SNITER; ++k) {
       const int i = tid + k * NTHREADS;
       srow[k] = i / QPR;
       sq[k] = i % QPR;
       ga[k] = max(0, min(gx0 + 16 * sq[k], pitchE - 16));
     }
 
-    auto issue = [&](uint4 (&pre)[NITER], int tile) {
+    // steady-state staging: the TILE_Y new rows of tile `tile` are
+    // chain-relative rows tile*TILE_Y + 2 .. + 10
+    uint4 preA[SNITER], preB[SNITER];
+    auto issue = [&](uint4 (&pre)[SNITER], int tile) {
 #pragma unroll
-      for (int k = 0; k < NITER; ++k) {
-        if (k == NITER - 1 && ITEMS % NTHREADS != 0 &&
-            tid >= ITEMS - k * NTHREADS)
+      for (int k = 0; k < SNITER; ++k) {
+        if (k == SNITER - 1 && SITEMS % NTHREADS != 0 &&
+            tid >= SITEMS - k * NTHREADS)
           break;
-        const int gy =
-            min(max(tyBase + tile * TILE_Y - 1 + srow[k], 0), bufh - 1);
+        const int rr = tile * TILE_Y + 2 + srow[k];
+        const int gy = min(tyBase - 1 + rr, bufh - 1);
         pre[k] = *reinterpret_cast<const uint4*>(src + (size_t)gy * pitchE +
                                                  ga[k]);
       }
     };
-    auto writeB = [&](const uint4 (&pre)[NITER], int buf) {
+    auto writeB = [&](const uint4 (&pre)[SNITER], int tile) {
 #pragma unroll
-      for (int k = 0; k < NITER; ++k) {
-        if (k == NITER - 1 && ITEMS % NTHREADS != 0 &&
-            tid >= ITEMS - k * NTHREADS)
+      for (int k = 0; k < SNITER; ++k) {
+        if (k == SNITER - 1 && SITEMS % NTHREADS != 0 &&
+            tid >= SITEMS - k * NTHREADS)
           break;
-        *reinterpret_cast<uint4*>(&lds[buf][srow[k]][16 * sq[k]]) = pre[k];
+        const int rr = tile * TILE_Y + 2 + srow[k];
+        *reinterpret_cast<uint4*>(&lds[slot(rr)][16 * sq[k]]) = pre[k];
       }
     };
 
     const int tx = tid & 63;
     const int tyy = tid >> 6;
 
-    auto compute_pass = [&](int t, int buf, int cp) {
+    auto compute_pass = [&](int t, int cp) {
       const int ty0 = tyBase + t * TILE_Y;
+      const int rbase = t * TILE_Y;  // chain-relative first input row
       const int txc = tx + 64 * cp;
       const int ex = etx0 + VEC * txc;
-      // 12-byte window per input row: bytes [4txc+12, 4txc+24) of the
-      // staged row; out px p uses local bytes (3+p, 4+p, 5+p)
       unsigned d[4][3];
 #pragma unroll
       for (int rw = 0; rw < 4; ++rw) {
         const unsigned* rp = reinterpret_cast<const unsigned*>(
-            &lds[buf][2 * tyy + rw][4 * txc + 12]);
+            &lds[slot(rbase + 2 * tyy + rw)][4 * txc + 12]);
         d[rw][0] = rp[0];
         d[rw][1] = rp[1];
         d[rw][2] = rp[2];
       }
-      // packed 4-byte pixel groups per (row, p): taps at local bytes
-      // 3+p..5+p; 4th byte is junk (its weight byte is 0)
       unsigned pk[4][4];
 #pragma unroll
       for (int rw = 0; rw < 4; ++rw) {
@@ -482,7 +494,6 @@ __global__ __launch_bounds__(NTHREADS) void conv_u8dot_k(
             const unsigned sel =
                 (unsigned)(3 + p) | ((unsigned)(4 + p) << 8) |
                 ((unsigned)(5 + p) << 16);
-            // {S0:high, S1:low}: byte index 0-3 = S1, 4-7 = S0
             pk[rw][p] = __builtin_amdgcn_perm(d[rw][1], d[rw][0], sel);
           } else {
             const unsigned sel = 2u | (3u << 8) | (4u << 16);
@@ -497,13 +508,13 @@ __global__ __launch_bounds__(NTHREADS) void conv_u8dot_k(
         unsigned qb = 0;
 #pragma unroll
         for (int p = 0; p < 4; ++p) {
-          unsigned s = half;
+          unsigned sacc = half;
 #pragma unroll
           for (int rw = 0; rw < 3; ++rw)
-            s = __builtin_amdgcn_udot4(pk[j + rw][p], wrow[rw], s, false);
-          unsigned q = s >> shift;
-          // half-even: s already includes +half, so a tie is (s&mask)==0
-          q -= ((s & mask) == 0u) & q & 1u;
+            sacc = __builtin_amdgcn_udot4(pk[j + rw][p], wrow[rw], sacc, false);
+          unsigned q = sacc >> shift;
+          // half-even: sacc already includes +half → a tie is (s&mask)==0
+          q -= ((sacc & mask) == 0u) & q & 1u;
           qb |= q << (8 * p);
         }
         unsigned char* out = dst + (size_t)y * pitchE + ex;
@@ -517,28 +528,38 @@ __global__ __launch_bounds__(NTHREADS) void conv_u8dot_k(
         }
       }
     };
-
-    auto compute_store = [&](int t, int buf) {
+    auto compute_store = [&](int t) {
 #pragma unroll
-      for (int cp = 0; cp < CP; ++cp) compute_pass(t, buf, cp);
+      for (int cp = 0; cp < CP; ++cp) compute_pass(t, cp);
     };
 
-    uint4 preA[NITER], preB[NITER];
-    issue(preA, 0);
-    writeB(preA, 0);
-    if (nt > 1) issue(preA, 1);
+    // prologue: stage all ROWS_IN rows of tile 0 (rr 0..9, top clamp)
+#pragma unroll
+    for (int k = 0; k < PNITER; ++k) {
+      if (k == PNITER - 1 && PITEMS % NTHREADS != 0 &&
+          tid >= PITEMS - k * NTHREADS)
+        break;
+      const int i = tid + k * NTHREADS;
+      const int rr = i / QPR;
+      const int q = i % QPR;
+      const int gaq = max(0, min(gx0 + 16 * q, pitchE - 16));
+      const int gy = min(max(tyBase - 1 + rr, 0), bufh - 1);
+      *reinterpret_cast<uint4*>(&lds[slot(rr)][16 * q]) =
+          *reinterpret_cast<const uint4*>(src + (size_t)gy * pitchE + gaq);
+    }
+    if (nt > 1) issue(preA, 1);  // tile 1's new rows = chain rows 10..17
     __syncthreads();
 
     for (int tt = 0; tt < nt; tt += 2) {
       if (tt + 2 < nt) issue(preB, tt + 2);
-      compute_store(tt, 0);
-      if (tt + 1 < nt) writeB(preA, 1);
+      compute_store(tt);
+      if (tt + 1 < nt) writeB(preA, tt + 1);
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();
       if (tt + 1 >= nt) break;
       if (tt + 3 < nt) issue(preA, tt + 3);
-      compute_store(tt + 1, 1);
-      if (tt + 2 < nt) writeB(preB, 0);
+      compute_store(tt + 1);
+      if (tt + 2 < nt) writeB(preB, tt + 2);
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();
     }
