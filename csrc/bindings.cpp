@@ -4,7 +4,6 @@
 // buffers are torch CUDA tensors (their allocator), the iteration loop,
 // streams and RCCL communicator live in C++ (csrc/engine.hip).
 
-#include <ATen/cuda/CUDAContext.h>
 #include <torch/extension.h>
 
 #include "common.hpp"
@@ -58,9 +57,14 @@ void conv_region(torch::Tensor src, torch::Tensor dst, torch::Tensor wgt,
   g.S = (int)channels;
   pconv::Region r{(int)region[0], (int)region[1], (int)region[2],
                   (int)region[3]};
-  auto stream = at::cuda::getCurrentCUDAStream();
+  // Test-only entry point: launch on the null stream, which serializes
+  // with torch's default stream on ROCm (torch work on the tensors has
+  // completed before and after this call). No torch stream API — this
+  // file builds HIP-native, with no hipify pass (BASELINE.json:5 "no
+  // CUDA-compat shims").
   pconv::launch_conv(dt, (int)channels, (int)k, src.data_ptr(), dst.data_ptr(),
-                     wgt.data_ptr<float>(), g, r, use_mfma, stream.stream());
+                     wgt.data_ptr<float>(), g, r, use_mfma,
+                     (hipStream_t) nullptr);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -112,11 +116,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     TORCH_CHECK(A.sizes() == torch::IntArrayRef({16, 32}));
     TORCH_CHECK(B.sizes() == torch::IntArrayRef({32, 16}));
     auto D = torch::zeros({16, 16}, A.options());
-    auto stream = at::cuda::getCurrentCUDAStream();
     pconv::launch_mfma_probe(A.contiguous().data_ptr<float>(),
                              B.contiguous().data_ptr<float>(),
-                             D.data_ptr<float>(), stream.stream());
-    HIP_CHECK(hipStreamSynchronize(stream.stream()));
+                             D.data_ptr<float>(), (hipStream_t) nullptr);
+    HIP_CHECK(hipStreamSynchronize((hipStream_t) nullptr));
     return D;
   });
 
