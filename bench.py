@@ -33,7 +33,7 @@ from parallel_convolution_amd.models.filters import (  # noqa: E402
 from parallel_convolution_amd.parallel import bootstrap  # noqa: E402
 from parallel_convolution_amd.parallel.decomp import halo_for_radius, make_blocks  # noqa: E402
 from parallel_convolution_amd.spec import DType, ProblemSpec  # noqa: E402
-from parallel_convolution_amd.utils.rawio import synthetic_image  # noqa: E402
+from parallel_convolution_amd.utils.rawio import synthetic_block, synthetic_image  # noqa: E402
 
 PRESETS = {
     # BASELINE.json:6-12 configs 2-5 (config 1 is the CPU plumbing test)
@@ -45,7 +45,7 @@ PRESETS = {
 }
 
 
-def verify_distributed(rank: int, world: int) -> None:
+def verify_distributed(rank: int, world: int) -> bool:
     from parallel_convolution_amd.hip_engine import HipEngine
     from parallel_convolution_amd.ops.reference import run_reference
     from parallel_convolution_amd.parallel.decomp import halo_for_radius
@@ -59,6 +59,7 @@ def verify_distributed(rank: int, world: int) -> None:
     veng.load_block(0, img[mine.y0 : mine.y0 + mine.h, mine.x0 : mine.x0 + mine.w])
     veng.run()
     out = torch.from_numpy(veng.store_block(0))
+    ok = True
     # gather blocks to rank 0 over the gloo control group
     if rank == 0:
         full = np.zeros((512, 512), dtype=np.uint8)
@@ -79,9 +80,14 @@ def verify_distributed(rank: int, world: int) -> None:
         )
     else:
         dist.send(out, dst=0)
+    # every rank learns the verdict (consistent exit codes on FAIL)
+    okt = torch.tensor([1 if ok else 0], dtype=torch.int32)
+    dist.broadcast(okt, src=0)
+    ok = bool(okt.item())
     dist.barrier()
     del veng
     torch.cuda.empty_cache()
+    return ok
 
 
 def main() -> int:
@@ -166,16 +172,26 @@ def main() -> int:
     # 6 iters, RCCL exchange) before the timed run: the oracle comparison
     # lands in the bench log so a scaling number is never quoted off a
     # silently-broken exchange. Costs ~1 s; --no-verify skips.
+    verified = None  # None = not run, True/False = oracle verdict
     if world > 1 and args.backend == "hip" and not args.no_verify:
         try:
-            verify_distributed(rank, world)
-        except Exception as e:  # the check must never break the bench
-            print(f"[verify] skipped ({type(e).__name__}: {e})",
+            verified = verify_distributed(rank, world)
+        except Exception as e:  # a crashed check is reported, not hidden
+            print(f"[verify] errored ({type(e).__name__}: {e})",
                   file=sys.stderr, flush=True)
+            verified = False
 
-    # synthetic random-pixel block data (BASELINE north-star), per-rank seed
+    # synthetic random-pixel block data (BASELINE north-star): every rank
+    # generates its window of the SAME global image (world-size-invariant
+    # counter-hash), so outputs are comparable across GPU counts
     for i, bs in enumerate(my_blocks):
-        eng.load_block(i, synthetic_image(bs.h, bs.w, spec.channels, spec.dtype, seed=1234 + bs.rank))
+        eng.load_block(
+            i,
+            synthetic_block(
+                spec.height, spec.width, spec.channels, spec.dtype,
+                bs.y0, bs.x0, bs.h, bs.w, seed=1234,
+            ),
+        )
 
     def dev_sync():
         if args.backend == "hip":
@@ -217,8 +233,9 @@ def main() -> int:
             "scaling": "strong",
             "vs_baseline": None,
             "dtype": spec.dtype.value,
-            "data": "synthetic random-pixel image, seeded per rank",
+            "data": "synthetic random-pixel image (world-size-invariant seeded hash)",
             "backend": args.backend,
+            "verified": verified,
             "config": {
                 "model": f"iterconv-{spec.k}x{spec.k}-"
                 f"{'gray' if spec.channels == 1 else 'rgb'}-{spec.dtype.value}",
@@ -243,7 +260,10 @@ def main() -> int:
 
     if world > 1:
         bootstrap.finalize()
-    return 0
+    # A failed pre-bench oracle check invalidates the headline number:
+    # the JSON above records verified=false AND the process exits non-zero
+    # so a broken exchange can never produce quotable numbers.
+    return 0 if verified is not False else 1
 
 
 if __name__ == "__main__":

@@ -9,6 +9,12 @@ WARMUP=${2:-10}
 PRESET=${3:-headline}
 cd "$(dirname "$0")/.."
 OUT=bench/scaling_$(date +%Y%m%d_%H%M%S).jsonl
+# free ephemeral rendezvous port (concurrent runs on one host must not collide)
+PORT=${PCONV_MASTER_PORT:-$(python - <<'EOF'
+import socket
+s = socket.socket(); s.bind(("127.0.0.1", 0)); print(s.getsockname()[1]); s.close()
+EOF
+)}
 for N in 1 2 4 8; do
   AVAIL=$(python -c "import torch; print(torch.cuda.device_count())")
   if [ "$N" -gt "$AVAIL" ]; then
@@ -20,7 +26,7 @@ for N in 1 2 4 8; do
       --preset "$PRESET" --json-out "$OUT"
   else
     python -m torch.distributed.run --nnodes=1 --nproc-per-node="$N" \
-      --master-addr 127.0.0.1 --master-port 29531 \
+      --master-addr 127.0.0.1 --master-port "$PORT" \
       bench.py --gpus "$N" --steps "$STEPS" --warmup "$WARMUP" \
       --preset "$PRESET" --json-out "$OUT"
   fi

@@ -57,6 +57,8 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--resume", action="store_true", help="resume from checkpoint")
     p.add_argument("--watchdog", type=int, default=0, metavar="SEC",
                    help="abort if an engine segment exceeds SEC seconds")
+    p.add_argument("--master-port", type=int, default=0, metavar="PORT",
+                   help="rendezvous port for --gpus N (0 = pick a free one)")
     p.add_argument("--metrics-out", default=None, help="append JSON run record here")
     p.add_argument("--phase-timing", action="store_true",
                    help="per-phase (exchange/compute) hipEvent timing in the record")
@@ -68,15 +70,26 @@ def main(argv=None) -> int:
     args = build_parser().parse_args(argv)
     world_env = bootstrap.env_world()
     if args.gpus > 1 and world_env == 1:
-        # re-exec under torchrun, one rank per GPU
+        # re-exec under torchrun, one rank per GPU; rendezvous on a free
+        # ephemeral port by default so concurrent runs on one host never
+        # collide (--master-port pins it when reproducibility matters)
+        port = args.master_port or _free_port()
         cmd = [
             sys.executable, "-m", "torch.distributed.run",
             "--nnodes=1", f"--nproc-per-node={args.gpus}",
-            "--master-addr=127.0.0.1", "--master-port=29531",
+            "--master-addr=127.0.0.1", f"--master-port={port}",
             "-m", "parallel_convolution_amd",
         ] + (argv if argv is not None else sys.argv[1:])
         return subprocess.call(cmd)
     return run_rank(args)
+
+
+def _free_port() -> int:
+    import socket
+
+    with socket.socket(socket.AF_INET, socket.SOCK_STREAM) as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
 
 
 def _alarm_handler(signum, frame):
@@ -138,7 +151,12 @@ def run_rank(args) -> int:
             ckpt_raw, spec.height, spec.width, S, spec.dtype, my.y0, my.x0, my.h, my.w
         )
     elif args.image == "synthetic":
-        img_block = rawio.synthetic_image(my.h, my.w, S, spec.dtype, seed=args.seed + rank)
+        # world-size-invariant: each rank hashes its window of the same
+        # seeded global image, so outputs match across GPU counts
+        img_block = rawio.synthetic_block(
+            spec.height, spec.width, S, spec.dtype,
+            my.y0, my.x0, my.h, my.w, seed=args.seed,
+        )
     else:
         img_block = rawio.read_block(
             args.image, spec.height, spec.width, S, spec.dtype, my.y0, my.x0, my.h, my.w
@@ -182,11 +200,15 @@ def run_rank(args) -> int:
             return blk.store()
 
     # --- iterate with convergence/checkpoint segments ------------------
-    seg_points = []
-    for every in (args.converge_every, args.checkpoint_every):
-        if every and every > 0:
-            seg_points.append(every)
-    seg = min(seg_points) if seg_points else max(args.iters - iters_done, 0)
+    # Each segment runs to the NEXT upcoming converge/checkpoint boundary
+    # (not a fixed min interval): --converge-every 2 --checkpoint-every 3
+    # stops at 2,3,4,6,... so no boundary is ever skipped.
+    def next_stop(done: int) -> int:
+        stops = [args.iters]
+        for every in (args.converge_every, args.checkpoint_every):
+            if every and every > 0:
+                stops.append((done // every + 1) * every)
+        return min(min(stops), args.iters)
 
     if world > 1:
         dist.barrier()
@@ -195,7 +217,7 @@ def run_rank(args) -> int:
     t0 = time.perf_counter()
     converged = False
     while iters_done < args.iters and not converged:
-        n = min(seg if seg > 0 else args.iters - iters_done, args.iters - iters_done)
+        n = next_stop(iters_done) - iters_done
         if args.watchdog > 0:
             signal.signal(signal.SIGALRM, _alarm_handler)
             signal.alarm(args.watchdog)

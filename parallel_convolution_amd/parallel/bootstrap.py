@@ -36,7 +36,11 @@ def init_control_group(backend: str = "gloo", port: Optional[int] = None) -> Non
     if dist.is_initialized():
         return
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-    os.environ.setdefault("MASTER_PORT", str(port or 29531))
+    # Under torchrun MASTER_PORT is always set; this fallback (world of 1,
+    # or hand-launched ranks) prefers PCONV_MASTER_PORT over the default.
+    os.environ.setdefault(
+        "MASTER_PORT", str(port or os.environ.get("PCONV_MASTER_PORT") or 29531)
+    )
     os.environ.setdefault("RANK", "0")
     os.environ.setdefault("WORLD_SIZE", "1")
     dist.init_process_group(backend=backend)

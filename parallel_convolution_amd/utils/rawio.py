@@ -125,23 +125,68 @@ def write_block(
         os.close(fd)
 
 
+_U64 = np.uint64
+
+
+def _splitmix64(x: np.ndarray) -> np.ndarray:
+    """Vectorized splitmix64 finalizer (counter-based RNG: value depends
+    only on the input word, so any sub-rectangle of the global image can
+    be generated independently)."""
+    z = (x + _U64(0x9E3779B97F4A7C15)).astype(np.uint64)
+    z ^= z >> _U64(30)
+    z *= _U64(0xBF58476D1CE4E5B9)
+    z ^= z >> _U64(27)
+    z *= _U64(0x94D049BB133111EB)
+    z ^= z >> _U64(31)
+    return z
+
+
+def synthetic_block(
+    height: int,
+    width: int,
+    channels: int,
+    dtype: DType,
+    y0: int,
+    x0: int,
+    h: int,
+    w: int,
+    seed: int = 1234,
+) -> np.ndarray:
+    """One (h, w)-pixel window of the seeded synthetic global image.
+
+    World-size-INVARIANT: pixel (y, x, c) of the global height×width frame
+    has the same value no matter which rank generates it or how the frame
+    is decomposed, so 1-GPU and N-GPU runs on `synthetic` input produce
+    comparable outputs (counter-based hash of the global element index).
+    Returns (h, w*channels).
+    """
+    wE = w * channels
+    row_e = width * channels
+    out = np.empty((h, wE), dtype=dtype.numpy_dtype)
+    sk = _splitmix64(np.array([seed & 0xFFFFFFFFFFFFFFFF], dtype=np.uint64))[0]
+    base_x = np.arange(x0 * channels, x0 * channels + wE, dtype=np.uint64)
+    # row slabs bound the u64 intermediate to ~1 GB (the 65536² config
+    # would otherwise need a 34 GB temporary)
+    slab = max(1, (1 << 27) // max(wE, 1))
+    for r0 in range(0, h, slab):
+        r1 = min(r0 + slab, h)
+        ys = np.arange(y0 + r0, y0 + r1, dtype=np.uint64) * _U64(row_e)
+        hv = _splitmix64((ys[:, None] + base_x[None, :]) ^ sk)
+        if dtype == DType.U8:
+            out[r0:r1] = (hv & _U64(0xFF)).astype(np.uint8)
+        else:
+            # floats in [0, 256) — same dynamic range as u8
+            out[r0:r1] = (
+                (hv >> _U64(40)).astype(np.float32) * np.float32(256.0 / (1 << 24))
+            ).astype(dtype.numpy_dtype)
+    return out
+
+
 def synthetic_image(
     height: int, width: int, channels: int, dtype: DType, seed: int = 1234
 ) -> np.ndarray:
     """Seeded random-pixel frame (BASELINE.json north-star: benchmarks run
     'on synthetic random-pixel images'); (height, width*channels)."""
-    rng = np.random.default_rng(seed)
-    shape = (height, width * channels)
-    if dtype == DType.U8:
-        return rng.integers(0, 256, size=shape, dtype=np.uint8)
-    # float images in [0, 256) — same dynamic range as u8; generated in
-    # row slabs so the fp32 intermediate never exceeds ~1 GB (the 65536²
-    # fp16 config would otherwise need a 17 GB temporary)
-    out = np.empty(shape, dtype=dtype.numpy_dtype)
-    slab = max(1, (1 << 28) // max(shape[1], 1))
-    for y0 in range(0, height, slab):
-        y1 = min(y0 + slab, height)
-        out[y0:y1] = (rng.random((y1 - y0, shape[1]), dtype=np.float32) * 256).astype(
-            dtype.numpy_dtype
-        )
-    return out
+    return synthetic_block(
+        height, width, channels, dtype, 0, 0, height, width, seed=seed
+    )

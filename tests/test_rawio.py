@@ -46,3 +46,31 @@ def test_block_write_reassembles(tmp_path):
         )
     back = rawio.read_image(p, H, W, C, DType.F32)
     np.testing.assert_array_equal(img, back)
+
+
+def test_synthetic_block_world_size_invariant():
+    """Any decomposition of the synthetic image reproduces the exact same
+    global pixels (per-rank seeding made world sizes
+    incomparable; the counter-hash generator fixes that by construction)."""
+    for dt in (DType.U8, DType.F16, DType.F32):
+        img = rawio.synthetic_image(48, 56, 1, dt, seed=11)
+        for p in (2, 4, 8):
+            for b in make_blocks(48, 56, p, 1):
+                blk = rawio.synthetic_block(
+                    48, 56, 1, dt, b.y0, b.x0, b.h, b.w, seed=11
+                )
+                np.testing.assert_array_equal(
+                    blk, img[b.y0 : b.y0 + b.h, b.x0 : b.x0 + b.w]
+                )
+
+
+def test_synthetic_block_rgb_window():
+    img = rawio.synthetic_image(32, 40, 3, DType.U8, seed=5)
+    blk = rawio.synthetic_block(32, 40, 3, DType.U8, 4, 6, 12, 20, seed=5)
+    np.testing.assert_array_equal(blk, img[4:16, 18:78])
+
+
+def test_synthetic_values_change_with_seed():
+    a = rawio.synthetic_image(16, 16, 1, DType.U8, seed=1)
+    b = rawio.synthetic_image(16, 16, 1, DType.U8, seed=2)
+    assert not np.array_equal(a, b)
