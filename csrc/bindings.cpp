@@ -75,10 +75,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
 
   py::class_<pconv::ConvEngine>(m, "ConvEngine")
       .def(py::init<int, int, const std::vector<float>&, int, bool,
-                    std::vector<int>>(),
+                    std::vector<int>, std::vector<int>>(),
            py::arg("dtype"), py::arg("k"), py::arg("filter"),
            py::arg("channels"), py::arg("use_mfma") = false,
-           py::arg("dot_weights") = std::vector<int>())
+           py::arg("dot_weights") = std::vector<int>(),
+           py::arg("sep_weights") = std::vector<int>())
       .def("add_block",
            [](pconv::ConvEngine& e, torch::Tensor src, torch::Tensor dst,
               int h, int wPx, int haloPx, int S, std::vector<int> nbr) {

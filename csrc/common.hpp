@@ -68,6 +68,14 @@ void launch_conv_u8dot(const void* src, void* dst, const unsigned* w9,
                        int shift, const BlockGeom& g, const Region& r,
                        hipStream_t stream);
 
+// Separable integer u8 fast path (k in {3,5}, S=1, non-negative
+// separable dyadic filter w = outer(b, a)/2^shift with
+// 255*sum(a)*sum(b) + 2^(shift-1) <= 65535): exact vs the fp32
+// contract. sep = [shift, a0..a4, b0..b4] (unused taps zero).
+void launch_conv_u8sep(const void* src, void* dst, const int* sep, int k,
+                       const BlockGeom& g, const Region& r,
+                       hipStream_t stream);
+
 // Fused integer-dot double iteration (u8 mid image in LDS)
 void launch_conv_u8dot2(const void* src, void* dst, const unsigned* w9,
                         int shift, const BlockGeom& g, const Region& rm,

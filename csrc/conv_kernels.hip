@@ -566,6 +566,261 @@ __global__ __launch_bounds__(NTHREADS) void conv_u8dot_k(
   }
 }
 
+// ---- separable integer u8 kernel (headline fast path, round 2) ------------
+//
+// For u8 grayscale with a NON-NEGATIVE SEPARABLE dyadic filter
+// w = outer(b, a)/2^shift (integer a_i, b_j >= 0, a_i <= 255,
+// 255*sum(a)*sum(b) + 2^(shift-1) <= 65535 — both mandated Gaussian
+// blurs qualify) the iteration factorizes:
+//   * horizontal pass  h(y,x) = sum_i a_i p(y, x+i-R)  via v_dot4_u32_u8,
+//     computed ONCE per input row and reused by all 2R+1 output rows that
+//     read it (the non-separable dot kernel redoes all K*K taps per
+//     output row);
+//   * vertical pass on u16 PAIRS — two pixels per dword — with
+//     v_pk_mul_lo_u16 products and plain u32 adds (SWAR-safe: every
+//     partial sum < 2^16 by the eligibility bound, so no carry crosses
+//     the lane boundary);
+//   * exact round-half-even in 4 SWAR ops:  with C = half-1 folded into
+//     the accumulator init,  t = (acc>>shift) & 1 per lane,  q =
+//     (acc+t)>>shift  (q<=255 so byte 0 resp. 2 of the result dword IS
+//     the quantized pixel — one v_perm packs 4 outputs).
+// Instructions/pixel drop ~35 % vs conv_u8dot_k; the headline kernel is
+// issue-bound, not BW-bound (profiles/r01), so that is the lever
+// (VERDICT.md r01 next-round #1). Exactness argument is conv_u8dot_k's:
+// all integer sums < 2^16 < 2^24 are exact in the fp32 contract and
+// integer RNE == fp32 RNE on dyadic values (hip_engine.u8sep_weights).
+//
+// Tile: 1024 output elements (4 px/lane, 256 threads all along x) × 8
+// rows, ring-staged chain like conv_u8dot_k (8 new rows per tile, 2-deep
+// prefetch pipeline); the 2R boundary h-rows are carried in registers
+// across chain tiles so steady state computes exactly one horizontal
+// pass per input row.
+
+typedef unsigned short u16x2v __attribute__((ext_vector_type(2)));
+
+static __device__ __forceinline__ unsigned pk_mul_u16(unsigned a, unsigned b) {
+  u16x2v r = __builtin_bit_cast(u16x2v, a) * __builtin_bit_cast(u16x2v, b);
+  return __builtin_bit_cast(unsigned, r);
+}
+
+// second launch_bounds arg: 4 workgroups/CU is the LDS-bound occupancy
+// ceiling (34.3 KB of 160 KB each), so give the allocator the full
+// 512/(4 waves) = 128-VGPR budget — the default heuristic picked 97 and
+// spilled the prefetch registers to scratch, defeating the pipeline.
+template <int R>
+__global__ __launch_bounds__(NTHREADS, 4) void conv_u8sep_k(
+    const unsigned char* __restrict__ src, unsigned char* __restrict__ dst,
+    unsigned wA, unsigned a4, unsigned bvp0, unsigned bvp1, unsigned bvp2,
+    unsigned bvp3, unsigned bvp4, int shift, int pitchE, int bufh, int ex0,
+    int ex1, int py0, int py1, int abase, int tilesX, int chainLen,
+    int nChains) {
+  constexpr int K = 2 * R + 1;
+  constexpr int TXE = 1024;  // output elements per tile (4 per lane)
+  constexpr int ROWS_IN = TILE_Y + 2 * R;
+  constexpr int RING = 32;  // > ROWS_IN + 2*TILE_Y in flight; pow2 slot
+  constexpr int WINB = TXE + 32;
+  constexpr int PITCHB = WINB + 16;
+  constexpr int QPR = WINB / 16;
+  constexpr int SITEMS = TILE_Y * QPR;
+  constexpr int PITEMS = ROWS_IN * QPR;
+  constexpr int SNITER = (SITEMS + NTHREADS - 1) / NTHREADS;
+  constexpr int PNITER = (PITEMS + NTHREADS - 1) / NTHREADS;
+
+  __shared__ unsigned char lds[RING][PITCHB];
+
+  const int tid = threadIdx.x;
+  const unsigned bvp[5] = {bvp0, bvp1, bvp2, bvp3, bvp4};
+  // round-half-even constant (half-1 per u16 lane), folded into acc init
+  const unsigned roundC = ((1u << (shift - 1)) - 1u) * 0x00010001u;
+
+  const int nWork = tilesX * nChains;
+  for (int work = blockIdx.x; work < nWork; work += gridDim.x) {
+    const int bx = work % tilesX;
+    const int bc = work / tilesX;
+    const int etx0 = abase + TXE * bx;
+    const int gx0 = etx0 - 16;
+    const int tyBase = py0 + TILE_Y * chainLen * bc;
+    const int nt = min(chainLen, (py1 - tyBase + TILE_Y - 1) / TILE_Y);
+
+    // ring slot of chain-relative input row rr (absolute tyBase - R + rr)
+    auto slot = [&](int rr) { return rr & (RING - 1); };
+
+    int srow[SNITER], sq[SNITER], ga[SNITER];
+#pragma unroll
+    for (int k = 0; k < SNITER; ++k) {
+      const int i = tid + k * NTHREADS;
+      srow[k] = i / QPR;
+      sq[k] = i % QPR;
+      ga[k] = max(0, min(gx0 + 16 * sq[k], pitchE - 16));
+    }
+
+    // prefetch registers are EXPLICIT SCALARS (not arrays): with
+    // SNITER=3 the array form failed SROA and lived in scratch — the
+    // spilled prefetch defeated the pipeline (seen in the ISA dump)
+    static_assert(SNITER == 3, "staging layout assumes 3 items/thread");
+    uint4 preA0, preA1, preA2, preB0, preB1, preB2;
+    auto issue1 = [&](uint4& pre, int k, int tile) {
+      if (tid + k * NTHREADS >= SITEMS) return;
+      const int rr = tile * TILE_Y + 2 * R + srow[k];
+      const int gy = min(tyBase - R + rr, bufh - 1);
+      pre = *reinterpret_cast<const uint4*>(src + (size_t)gy * pitchE + ga[k]);
+    };
+    auto write1 = [&](const uint4& pre, int k, int tile) {
+      if (tid + k * NTHREADS >= SITEMS) return;
+      const int rr = tile * TILE_Y + 2 * R + srow[k];
+      *reinterpret_cast<uint4*>(&lds[slot(rr)][16 * sq[k]]) = pre;
+    };
+    auto issueA = [&](int tile) {
+      issue1(preA0, 0, tile);
+      issue1(preA1, 1, tile);
+      issue1(preA2, 2, tile);
+    };
+    auto issueB = [&](int tile) {
+      issue1(preB0, 0, tile);
+      issue1(preB1, 1, tile);
+      issue1(preB2, 2, tile);
+    };
+    auto writeA = [&](int tile) {
+      write1(preA0, 0, tile);
+      write1(preA1, 1, tile);
+      write1(preA2, 2, tile);
+    };
+    auto writeBf = [&](int tile) {
+      write1(preB0, 0, tile);
+      write1(preB1, 1, tile);
+      write1(preB2, 2, tile);
+    };
+
+    // horizontal pass for chain row rr: h pairs (h01, h23) of this
+    // lane's 4 px — one v_perm window gather + one v_dot4 per pixel
+    // (K=5 adds the 5th tap via a zero-extended-byte second dot)
+    auto hrow = [&](int rr, unsigned* hp) {
+      const unsigned* rp =
+          reinterpret_cast<const unsigned*>(&lds[slot(rr)][4 * tid + 12]);
+      const unsigned d0 = rp[0], d1 = rp[1], d2 = rp[2];
+      unsigned h[4];
+      if constexpr (R == 1) {
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+          const unsigned pk =
+              (p < 3) ? __builtin_amdgcn_perm(
+                            d1, d0,
+                            (unsigned)(3 + p) | ((unsigned)(4 + p) << 8) |
+                                ((unsigned)(5 + p) << 16))
+                      : __builtin_amdgcn_perm(d2, d1, 2u | (3u << 8) | (4u << 16));
+          h[p] = __builtin_amdgcn_udot4(pk, wA, 0u, false);
+        }
+      } else {
+        // K=5 window bytes (relative to d0 base): pixel p covers 2+p..6+p
+        const unsigned pk0 =
+            __builtin_amdgcn_perm(d1, d0, 2u | (3u << 8) | (4u << 16) | (5u << 24));
+        const unsigned pk1 =
+            __builtin_amdgcn_perm(d1, d0, 3u | (4u << 8) | (5u << 16) | (6u << 24));
+        const unsigned pk2 = d1;  // bytes 4..7 exactly
+        const unsigned pk3 =
+            __builtin_amdgcn_perm(d2, d1, 1u | (2u << 8) | (3u << 16) | (4u << 24));
+        const unsigned b5[4] = {(d1 >> 16) & 0xffu, d1 >> 24, d2 & 0xffu,
+                                (d2 >> 8) & 0xffu};
+        h[0] = __builtin_amdgcn_udot4(pk0, wA, 0u, false);
+        h[1] = __builtin_amdgcn_udot4(pk1, wA, 0u, false);
+        h[2] = __builtin_amdgcn_udot4(pk2, wA, 0u, false);
+        h[3] = __builtin_amdgcn_udot4(pk3, wA, 0u, false);
+#pragma unroll
+        for (int p = 0; p < 4; ++p)
+          h[p] = __builtin_amdgcn_udot4(b5[p], a4, h[p], false);
+      }
+      hp[0] = __builtin_amdgcn_perm(h[1], h[0], 0x05040100u);
+      hp[1] = __builtin_amdgcn_perm(h[3], h[2], 0x05040100u);
+    };
+
+    const int ex = etx0 + VEC * tid;
+
+    // vertical pass + store for output row y; hw[0..K-1] are the h pairs
+    // of input rows y-R..y+R
+    auto vstore = [&](int y, const unsigned (*hw)[2]) {
+      unsigned acc0 = roundC, acc1 = roundC;
+#pragma unroll
+      for (int j = 0; j < K; ++j) {
+        acc0 += pk_mul_u16(bvp[j], hw[j][0]);
+        acc1 += pk_mul_u16(bvp[j], hw[j][1]);
+      }
+      // round-half-even: q = (acc + ((acc>>shift)&1)) >> shift per lane
+      acc0 += (acc0 >> shift) & 0x00010001u;
+      acc1 += (acc1 >> shift) & 0x00010001u;
+      acc0 >>= shift;
+      acc1 >>= shift;
+      const unsigned qb = __builtin_amdgcn_perm(acc1, acc0, 0x06040200u);
+      unsigned char* out = dst + (size_t)y * pitchE + ex;
+      if (ex >= ex0 && ex + VEC <= ex1) {
+        *reinterpret_cast<unsigned*>(out) = qb;
+      } else {
+#pragma unroll
+        for (int p = 0; p < VEC; ++p)
+          if (ex + p >= ex0 && ex + p < ex1)
+            out[p] = (unsigned char)((qb >> (8 * p)) & 0xffu);
+      }
+    };
+
+    unsigned hc[2 * R][2];  // carried boundary h rows (chain invariant)
+
+    auto compute_tile = [&](int t) {
+      unsigned hl[TILE_Y + 2 * R][2];
+#pragma unroll
+      for (int i = 0; i < 2 * R; ++i) {
+        hl[i][0] = hc[i][0];
+        hl[i][1] = hc[i][1];
+      }
+      const int ty0 = tyBase + t * TILE_Y;
+#pragma unroll
+      for (int j = 0; j < TILE_Y; ++j) {
+        hrow(t * TILE_Y + j + 2 * R, hl[j + 2 * R]);
+        const int y = ty0 + j;
+        if (y < py1) vstore(y, &hl[j]);
+      }
+#pragma unroll
+      for (int i = 0; i < 2 * R; ++i) {
+        hc[i][0] = hl[TILE_Y + i][0];
+        hc[i][1] = hl[TILE_Y + i][1];
+      }
+    };
+
+    // prologue: stage all ROWS_IN rows of tile 0 (rr 0..ROWS_IN-1)
+#pragma unroll
+    for (int k = 0; k < PNITER; ++k) {
+      if (k == PNITER - 1 && PITEMS % NTHREADS != 0 &&
+          tid >= PITEMS - k * NTHREADS)
+        break;
+      const int i = tid + k * NTHREADS;
+      const int rr = i / QPR;
+      const int q = i % QPR;
+      const int gaq = max(0, min(gx0 + 16 * q, pitchE - 16));
+      const int gy = min(max(tyBase - R + rr, 0), bufh - 1);
+      *reinterpret_cast<uint4*>(&lds[slot(rr)][16 * q]) =
+          *reinterpret_cast<const uint4*>(src + (size_t)gy * pitchE + gaq);
+    }
+    if (nt > 1) issueA(1);
+    __syncthreads();
+
+    // boundary h rows rr = 0..2R-1 seed the carry
+#pragma unroll
+    for (int i = 0; i < 2 * R; ++i) hrow(i, hc[i]);
+
+    for (int tt = 0; tt < nt; tt += 2) {
+      if (tt + 2 < nt) issueB(tt + 2);
+      compute_tile(tt);
+      if (tt + 1 < nt) writeA(tt + 1);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      if (tt + 1 >= nt) break;
+      if (tt + 3 < nt) issueA(tt + 3);
+      compute_tile(tt + 1);
+      if (tt + 2 < nt) writeBf(tt + 2);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+}
+
 // ---- fused integer-dot u8 double-iteration kernel -------------------------
 //
 // conv_u8dot_k x conv_tile2_k: two exact-integer iterations per launch
@@ -1398,6 +1653,31 @@ void launch_u8dot_impl(const void* src, void* dst, const unsigned* w9,
     launch_u8dot_cp<1>(src, dst, w9, shift, g, r, stream);
 }
 
+template <int R>
+void launch_u8sep_r(const void* src, void* dst, const int* sep,
+                    const BlockGeom& g, const Region& r, hipStream_t stream) {
+  // sep = [shift, a0..a4, b0..b4] (hip_engine.u8sep_weights layout)
+  const int TXE = 1024;
+  const int abase = (r.ex0 / 16) * 16;
+  const int tilesX = (r.ex1 - abase + TXE - 1) / TXE;
+  const int tilesY = (r.py1 - r.py0 + TILE_Y - 1) / TILE_Y;
+  const int chainLen = chain_for(tilesX, tilesY);
+  const int chains = (tilesY + chainLen - 1) / chainLen;
+  const int grid = grid_for((const void*)&conv_u8sep_k<R>, tilesX * chains);
+  const unsigned* a = reinterpret_cast<const unsigned*>(sep + 1);
+  const unsigned* b = reinterpret_cast<const unsigned*>(sep + 6);
+  const unsigned wA = a[0] | (a[1] << 8) | (a[2] << 16) |
+                      (R == 2 ? (a[3] << 24) : 0u);
+  const unsigned a4 = R == 2 ? a[4] : 0u;
+  unsigned bvp[5] = {};
+  for (int j = 0; j < 2 * R + 1; ++j) bvp[j] = b[j] * 0x00010001u;
+  hipLaunchKernelGGL((conv_u8sep_k<R>), dim3(grid), dim3(NTHREADS), 0, stream,
+                     (const unsigned char*)src, (unsigned char*)dst, wA, a4,
+                     bvp[0], bvp[1], bvp[2], bvp[3], bvp[4], sep[0], g.pitchE,
+                     g.bufh, r.ex0, r.ex1, r.py0, r.py1, abase, tilesX,
+                     chainLen, chains);
+}
+
 void launch_u8dot2_impl(const void* src, void* dst, const unsigned* w9,
                         int shift, const BlockGeom& g, const Region& rm,
                         const Region& ro, hipStream_t stream) {
@@ -1418,6 +1698,16 @@ void launch_u8dot2_impl(const void* src, void* dst, const unsigned* w9,
 }
 
 }  // namespace
+
+void launch_conv_u8sep(const void* src, void* dst, const int* sep, int k,
+                       const BlockGeom& g, const Region& r,
+                       hipStream_t stream) {
+  if (r.py1 <= r.py0 || r.ex1 <= r.ex0) return;
+  if (k == 3)
+    launch_u8sep_r<1>(src, dst, sep, g, r, stream);
+  else
+    launch_u8sep_r<2>(src, dst, sep, g, r, stream);
+}
 
 void launch_conv_u8dot2(const void* src, void* dst, const unsigned* w9,
                         int shift, const BlockGeom& g, const Region& rm,

@@ -59,11 +59,18 @@ struct DeviceBlock {
 class ConvEngine {
  public:
   ConvEngine(int dtype, int k, const std::vector<float>& filter, int channels,
-             bool useMfma = false, std::vector<int> dotW = {})
+             bool useMfma = false, std::vector<int> dotW = {},
+             std::vector<int> sepW = {})
       : dt_(static_cast<DType>(dtype)), k_(k), S_(channels), useMfma_(useMfma) {
     if (dotW.size() == 10 && dt_ == DType::U8 && k == 3 && channels == 1) {
       dotShift_ = dotW[0];
       for (int i = 0; i < 9; ++i) dotW_[i] = (unsigned)dotW[1 + i];
+    }
+    // separable integer path (round 2 headline): [shift, a0..a4, b0..b4]
+    if (sepW.size() == 11 && dt_ == DType::U8 && (k == 3 || k == 5) &&
+        channels == 1) {
+      sepOn_ = true;
+      for (int i = 0; i < 11; ++i) sepW_[i] = sepW[i];
     }
     if ((int)filter.size() != k * k) throw std::runtime_error("filter size != k*k");
     HIP_CHECK(hipStreamCreateWithFlags(&compute_, hipStreamNonBlocking));
@@ -198,7 +205,7 @@ class ConvEngine {
       if (phaseTiming_) HIP_CHECK(hipEventRecord(evPhase_[2], compute_));
       const bool canFuse =
           fusePairs_ && !useMfma_ && r >= 1 && r <= 2 && (S_ == 1 || S_ == 3);
-      const bool canDotFuse = dotFuse_ && dotShift_ > 0;
+      const bool canDotFuse = dotFuse_ && dotShift_ > 0 && !sepOn_;
       for (int t = t0; t < T;) {
         if ((canFuse || canDotFuse) && t + 1 < T) {
           // fused pair: iteration t into LDS mid, iteration t+1 to global
@@ -249,6 +256,7 @@ class ConvEngine {
   // which compute path run() will take (observability: lands in the
   // bench JSON so a measurement names the kernel that produced it)
   std::string kernel_path() const {
+    if (sepOn_) return "u8sep";
     if (dotShift_ > 0) return dotFuse_ ? "u8dot-fused" : "u8dot";
     if (useMfma_) return "mfma-f16";
     if (fusePairs_ && k_ / 2 >= 1 && k_ / 2 <= 2 && (S_ == 1 || S_ == 3))
@@ -306,7 +314,9 @@ class ConvEngine {
   // single-iteration launch: integer-dot u8 fast path when eligible,
   // else MFMA/VALU via launch_conv
   void launch_region(DeviceBlock& b, const Region& rg) {
-    if (dotShift_ > 0) {
+    if (sepOn_) {
+      launch_conv_u8sep(b.src, b.dst, sepW_, k_, b.g, rg, compute_);
+    } else if (dotShift_ > 0) {
       launch_conv_u8dot(b.src, b.dst, dotW_, dotShift_, b.g, rg, compute_);
     } else {
       launch_conv(dt_, S_, k_, b.src, b.dst, (const float*)wgt_, b.g, rg,
@@ -440,6 +450,8 @@ class ConvEngine {
   bool swapped_ = false;
   bool useMfma_ = false;
   bool fusePairs_ = false;
+  bool sepOn_ = false;
+  int sepW_[11] = {};
   int dotShift_ = 0;
   bool dotFuse_ = false;
   unsigned dotW_[9] = {};

@@ -53,6 +53,61 @@ def mfma_eligible(spec: ProblemSpec) -> bool:
     return spec.k == 5
 
 
+def u8sep_weights(spec: ProblemSpec):
+    """Integer taps [shift, a0..a4, b0..b4] for the separable integer path
+    (csrc conv_u8sep_k): k in {3,5} grayscale u8 with a non-negative
+    SEPARABLE dyadic filter w = outer(b, a)/2^shift (integer a, b >= 0,
+    a_i <= 255) whose worst-case accumulator fits u16:
+    255*sum(a)*sum(b) + 2^(shift-1) <= 65535. Both mandated Gaussian
+    blurs qualify; the horizontal pass is then shared across the 2R+1
+    output rows that read it (instructions/pixel).
+    Empty list = ineligible. PCONV_DISABLE_SEP=1 forces the next path."""
+    import math
+    import os
+
+    if os.environ.get("PCONV_DISABLE_SEP"):
+        return []
+    if spec.dtype != DType.U8 or spec.k not in (3, 5) or spec.channels != 1:
+        return []
+    w = spec.filter.astype("float64").reshape(spec.k, spec.k)
+    if (w < 0).any():
+        return []
+    for shift in range(1, 9):
+        n = w * (1 << shift)
+        if not (n == n.round()).all():
+            continue
+        n = n.round().astype("int64")
+        # integer rank-1 factorization n = outer(b, a)
+        sums = n.sum(axis=1)
+        i0 = int(sums.argmax())
+        if sums[i0] == 0:
+            return []
+        row = n[i0]
+        g = int(math.gcd(*[int(v) for v in row]))
+        if g == 0:
+            return []
+        a = row // g
+        j0 = int(a.argmax())
+        if a[j0] == 0:
+            return []
+        b = n[:, j0] // a[j0]
+        if (np.outer(b, a) != n).any():
+            return []
+        if a.max() > 255:
+            return []
+        if 255 * int(a.sum()) * int(b.sum()) + (1 << (shift - 1)) > 65535:
+            return []
+        pad = [0] * (5 - spec.k)
+        return (
+            [shift]
+            + [int(v) for v in a]
+            + pad
+            + [int(v) for v in b]
+            + pad
+        )
+    return []
+
+
 def u8dot_weights(spec: ProblemSpec):
     """Integer taps [shift, n0..n8] for the exact u8 dot-product path
     (csrc conv_u8dot_k): k=3 grayscale with a non-negative dyadic filter
@@ -117,6 +172,7 @@ class HipEngine:
         filt = [float(v) for v in spec.filter.flat]
         self.use_mfma = mfma_eligible(spec)
         self.dot_weights = u8dot_weights(spec)
+        self.sep_weights = u8sep_weights(spec)
         self.engine = self._C.ConvEngine(
             _DT_CODE[spec.dtype],
             spec.k,
@@ -124,6 +180,7 @@ class HipEngine:
             spec.channels,
             self.use_mfma,
             self.dot_weights,
+            self.sep_weights,
         )
         for bs in blocks:
             shape = (bs.buf_h, bs.buf_w_elems)

@@ -442,3 +442,56 @@ def test_fused_dot_byte_exact(p):
         del os.environ["PCONV_DOT_FUSE"]
     ref = run_reference(spec, img)
     np.testing.assert_array_equal(out, ref)
+
+
+# ---- separable integer path (round-2 headline kernel) ----------------------
+
+
+def test_sep_selected_for_headline_filter():
+    from parallel_convolution_amd.hip_engine import HipEngine
+    from parallel_convolution_amd.parallel.decomp import make_blocks
+
+    spec = ProblemSpec(64, 64, 1, DType.U8, GAUSSIAN_BLUR_3, 1)
+    eng = HipEngine(spec, make_blocks(64, 64, 1, 1))
+    assert eng.kernel_path() == "u8sep"
+    spec5 = ProblemSpec(64, 64, 1, DType.U8, GAUSSIAN_BLUR_5, 1)
+    eng5 = HipEngine(spec5, make_blocks(64, 64, 1, 1))
+    assert eng5.kernel_path() == "u8sep"
+
+
+@pytest.mark.parametrize("W,H,filt,iters", [
+    (512, 384, GAUSSIAN_BLUR_3, 10),
+    (1031, 517, GAUSSIAN_BLUR_3, 7),   # odd sizes, 1024-wide tile clipping
+    (512, 384, GAUSSIAN_BLUR_5, 6),    # k=5 separable path
+    (1031, 517, GAUSSIAN_BLUR_5, 5),
+    (900, 40, GAUSSIAN_BLUR_3, 3),     # fewer rows than one tile chain
+])
+def test_sep_matches_oracle(W, H, filt, iters):
+    spec = ProblemSpec(W, H, 1, DType.U8, filt, iters)
+    img = synthetic_image(H, W, 1, DType.U8)
+    out = _engine_mod().run_single_gpu(spec, img)
+    ref = run_reference(spec, img)
+    np.testing.assert_array_equal(out, ref)
+
+
+@pytest.mark.parametrize("p,filt", [(2, GAUSSIAN_BLUR_3), (8, GAUSSIAN_BLUR_3),
+                                    (4, GAUSSIAN_BLUR_5), (8, GAUSSIAN_BLUR_5)])
+def test_sep_multiblock_protocol(p, filt):
+    """Halo-deepened chunked exchange + sep kernel == oracle (the border
+    strip regions exercise narrow-region launches of the 1024-wide tile)."""
+    spec = ProblemSpec(160, 144, 1, DType.U8, filt, 20)
+    img = synthetic_image(144, 160, 1, DType.U8)
+    out = _engine_mod().run_local_blocks_gpu(spec, img, p)
+    ref = run_reference(spec, img)
+    np.testing.assert_array_equal(out, ref)
+
+
+def test_sep_vs_dot_identical(monkeypatch):
+    """A/B: the separable kernel and the non-separable dot kernel are
+    bit-identical on the headline filter (both exact integer RNE)."""
+    spec = ProblemSpec(640, 480, 1, DType.U8, GAUSSIAN_BLUR_3, 9)
+    img = synthetic_image(480, 640, 1, DType.U8, seed=23)
+    out_sep = _engine_mod().run_single_gpu(spec, img)
+    monkeypatch.setenv("PCONV_DISABLE_SEP", "1")
+    out_dot = _engine_mod().run_single_gpu(spec, img)
+    np.testing.assert_array_equal(out_sep, out_dot)

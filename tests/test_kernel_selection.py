@@ -62,3 +62,97 @@ def test_mfma_force(monkeypatch):
     assert mfma_eligible(spec(GAUSSIAN_BLUR_3))
     monkeypatch.setenv("PCONV_DISABLE_MFMA", "1")
     assert not mfma_eligible(spec(GAUSSIAN_BLUR_3))
+
+
+# ---- separable integer path (round 2 headline kernel) ----------------------
+
+from parallel_convolution_amd.hip_engine import u8sep_weights  # noqa: E402
+
+
+def test_sep_blur3_eligible():
+    assert u8sep_weights(spec(GAUSSIAN_BLUR_3)) == [
+        4, 1, 2, 1, 0, 0, 1, 2, 1, 0, 0
+    ]
+
+
+def test_sep_blur5_eligible():
+    w = u8sep_weights(spec(GAUSSIAN_BLUR_5))
+    assert w == [8, 1, 4, 6, 4, 1, 1, 4, 6, 4, 1]
+    # u16 accumulator bound holds
+    assert 255 * 16 * 16 + (1 << 7) <= 65535
+
+
+def test_sep_identity_eligible():
+    w = u8sep_weights(spec(IDENTITY_3))
+    assert w and w[0] >= 1
+
+
+def test_sep_ineligible_cases():
+    assert u8sep_weights(spec(EDGE_3)) == []        # negative taps
+    assert u8sep_weights(spec(SHARPEN_5)) == []     # negative taps
+    assert u8sep_weights(spec(GAUSSIAN_BLUR_3, c=3)) == []  # RGB
+    assert u8sep_weights(spec(GAUSSIAN_BLUR_3, dt=DType.F16)) == []
+    # dyadic but NOT separable: blur3 with a dented center stays dyadic
+    # (sum 15/16 <= 1) but is rank 2
+    f = GAUSSIAN_BLUR_3.copy()
+    f[1, 1] -= np.float32(1.0 / 16.0)
+    assert u8sep_weights(spec(f)) == []
+    assert u8dot_weights(spec(f)) != []  # ...but the dot path covers it
+
+
+def test_sep_disable_env(monkeypatch):
+    monkeypatch.setenv("PCONV_DISABLE_SEP", "1")
+    assert u8sep_weights(spec(GAUSSIAN_BLUR_3)) == []
+
+
+def _sep_simulate(img, w9, k):
+    """Simulate the conv_u8sep_k integer arithmetic (horizontal udot4,
+    u16 vertical SWAR, round-half-even fix) in numpy — the exactness
+    argument of csrc/conv_kernels.hip conv_u8sep_k, testable on CPU."""
+    shift = w9[0]
+    a = np.array(w9[1 : 1 + k], dtype=np.uint32)
+    b = np.array(w9[6 : 6 + k], dtype=np.uint32)
+    r = k // 2
+    p = np.pad(img.astype(np.uint32), r)
+    H, W = img.shape
+    # horizontal pass on all input rows
+    h = np.zeros((H + 2 * r, W), dtype=np.uint32)
+    for i in range(k):
+        h += a[i] * p[:, i : i + W]
+    assert h.max() < 1 << 16
+    # vertical pass + rounding
+    acc = np.full((H, W), (1 << (shift - 1)) - 1, dtype=np.uint32)
+    for j in range(k):
+        acc += b[j] * h[j : j + H, :]
+    assert acc.max() < 1 << 16
+    t = (acc >> shift) & 1
+    q = (acc + t) >> shift
+    assert q.max() <= 255
+    return q.astype(np.uint8)
+
+
+@pytest.mark.parametrize("filt,k", [(GAUSSIAN_BLUR_3, 3), (GAUSSIAN_BLUR_5, 5)])
+def test_sep_integer_arithmetic_matches_oracle(filt, k):
+    from parallel_convolution_amd.ops.reference import run_reference
+    from parallel_convolution_amd.utils.rawio import synthetic_image
+
+    s = ProblemSpec(96, 64, 1, DType.U8, filt, 1)
+    img = synthetic_image(64, 96, 1, DType.U8, seed=17)
+    w9 = u8sep_weights(s)
+    assert w9
+    got = _sep_simulate(img, w9, k)
+    ref = run_reference(s, img)
+    np.testing.assert_array_equal(got, ref)
+
+
+def test_sep_rounding_edge_values():
+    """All 256^3 vertical-window values would be slow; sweep the full
+    s-range for blur3 instead: every achievable integer sum s in
+    [0, 255*16] must round exactly like fp32 RNE."""
+    shift = 4
+    s = np.arange(0, 255 * 16 + 1, dtype=np.uint32)
+    v = s + (1 << (shift - 1)) - 1
+    t = (v >> shift) & 1
+    q = (v + t) >> shift
+    ref = np.rint(s.astype(np.float64) / (1 << shift)).astype(np.uint32)
+    np.testing.assert_array_equal(q, ref)
