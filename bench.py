@@ -139,7 +139,19 @@ def main() -> int:
 
         eng = HipEngine(spec, my_blocks, world_blocks=blocks, chunk_iters=args.chunk)
         if world > 1:
+            # a hung RCCL rendezvous must fail fast (and name itself),
+            # not eat the driver's whole timeout
+            import signal
+
+            def _to(signum, frame):
+                raise TimeoutError(
+                    f"rank {rank}: RCCL communicator init exceeded 180 s"
+                )
+
+            signal.signal(signal.SIGALRM, _to)
+            signal.alarm(180)
             eng.init_distributed(rank, world)
+            signal.alarm(0)
     else:
         from parallel_convolution_amd.cpu_engine import (
             CpuBlock,
@@ -220,6 +232,23 @@ def main() -> int:
     pixel_iters = float(spec.width) * spec.height * args.steps
     value = pixel_iters / elapsed / 1e9  # Gpixel-iters/s, whole job
 
+    # Post-bench phase diagnostic at world>1: a short
+    # phase-timed segment AFTER the timed region shows where each rank's
+    # iteration goes (exchange vs compute) if scaling is sub-linear.
+    # Costs one device sync per chunk — never inside the timed region.
+    phases = None
+    if world > 1 and args.backend == "hip" and hasattr(eng, "set_phase_timing"):
+        try:
+            eng.set_phase_timing(True)
+            eng.run(min(2 * eng.chunk_iters, 64))
+            dev_sync()
+            mine = eng.phase_times()
+            rows = [None] * world
+            dist.all_gather_object(rows, {"rank": rank, **mine})
+            phases = rows if rank == 0 else None
+        except Exception as e:
+            print(f"[phases] diagnostic failed: {e}", file=sys.stderr, flush=True)
+
     if rank == 0:
         result = {
             "metric": "Gpixel-iters/sec",
@@ -236,6 +265,7 @@ def main() -> int:
             "data": "synthetic random-pixel image (world-size-invariant seeded hash)",
             "backend": args.backend,
             "verified": verified,
+            "phases": phases,
             "config": {
                 "model": f"iterconv-{spec.k}x{spec.k}-"
                 f"{'gray' if spec.channels == 1 else 'rgb'}-{spec.dtype.value}",
