@@ -110,22 +110,39 @@ struct U8T {
     u.y = pack4(a + 4);
     *reinterpret_cast<uint2*>(p) = u;
   }
-  // fp16 staging for the MFMA path: 8 bytes -> 8 halves (u8 is exact in fp16)
+  // fp16 staging for the MFMA path: 8 bytes -> 8 halves (u8 is exact in
+  // fp16). Magic-number conversion: the u16 bit pattern (0x6400 | v) IS
+  // f16(1024 + v) for v in 0..255 (1024..1279 fits the 10-bit mantissa
+  // exactly), so ONE v_perm interleaving pixel bytes with 0x64 builds
+  // two halves at once, and one exact v_pk_sub_f16 of 1024 finishes —
+  // 1 VALU op/element where the per-element cvt chain cost ~3 (this
+  // kernel is issue-bound: r02 PMC, profiles/).
   using RawH = uint2;
   static __device__ __forceinline__ RawH loadRawH(const T* p) {
     return *reinterpret_cast<const uint2*>(p);
   }
   static __device__ __forceinline__ void cvtWriteH(_Float16* p, RawH v) {
-    // u8 -> f16 directly via v_cvt_f16_u16 (exact for 0..255): one cvt
-    // per element instead of the 2-op u8->f32->f16 chain
-    _Float16 o[8] __attribute__((aligned(16)));
-    const unsigned* d = reinterpret_cast<const unsigned*>(&v);
+    const unsigned magic = 0x64646464u;
+    const __half2 kBias =
+        __halves2half2(__ushort_as_half(0x6400), __ushort_as_half(0x6400));
+    unsigned o[4] __attribute__((aligned(16)));
+    const unsigned s01 = 0x05010400u, s23 = 0x05030402u;
+    o[0] = __builtin_amdgcn_perm(magic, v.x, s01);
+    o[1] = __builtin_amdgcn_perm(magic, v.x, s23);
+    o[2] = __builtin_amdgcn_perm(magic, v.y, s01);
+    o[3] = __builtin_amdgcn_perm(magic, v.y, s23);
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
-#pragma unroll
-      for (int b = 0; b < 4; ++b)
-        o[4 * i + b] = (_Float16)(unsigned short)((d[i] >> (8 * b)) & 0xffu);
+    for (int i = 0; i < 4; ++i) {
+      __half2 h = __hsub2(*reinterpret_cast<const __half2*>(&o[i]), kBias);
+      o[i] = *reinterpret_cast<unsigned*>(&h);
+    }
     *reinterpret_cast<uint4*>(p) = *reinterpret_cast<const uint4*>(o);
+  }
+  // scattered single-element quantize (MFMA transpose scratch): RNE +
+  // saturate in 2 ops via rint + v_cvt_pk_u8_f32
+  static __device__ __forceinline__ void storeQ1(T* p, float v) {
+    *p = (T)(__builtin_amdgcn_cvt_pk_u8_f32(__builtin_rintf(v), 0, 0u) &
+             0xffu);
   }
 };
 
@@ -182,6 +199,9 @@ struct F16T {
   }
   static __device__ __forceinline__ void cvtWriteH(_Float16* p, RawH v) {
     *reinterpret_cast<uint4*>(p) = v;
+  }
+  static __device__ __forceinline__ void storeQ1(T* p, float v) {
+    store1(p, v);
   }
 };
 
@@ -1344,8 +1364,13 @@ __global__ __launch_bounds__(M_NT) void conv_mfma_k(
     }
 
     const int ty0 = tyBase + t * M_TILE_Y;
-    // four 16-element output subtiles per wave (independent MFMA
-    // accumulator chains → latency hiding)
+    // four 16-element output subtiles per wave, computed in one phase so
+    // the A-fragment ds_reads of subtile i+1 and the MFMA chains of
+    // subtile i interleave (the old per-subtile compute→quantize order
+    // stalled on each accumulator readback right after its 5-chain: the
+    // r01 PMC showed SQ_WAIT_ANY at 48 % of wave cycles on this kernel);
+    // quantization happens in a second phase over all 16 accumulators.
+    f32x4v accs[4];
 #pragma unroll
     for (int sti = 0; sti < 4; ++sti) {
       const int st = 4 * wave + sti;
@@ -1358,15 +1383,18 @@ __global__ __launch_bounds__(M_NT) void conv_mfma_k(
             &lds[t & 1][row][(q ^ (row & 7)) * 8]);
         acc = __builtin_amdgcn_mfma_f32_16x16x32_f16(a, bfrag[dy], acc, 0, 0, 0);
       }
-      // quantize into the per-wave transpose scratch (accumulator layout:
-      // col = lane&15, row = 4*(lane>>4)+reg — scattered element stores
-      // go to LDS, the global stores below are contiguous 16-B runs)
+      accs[sti] = acc;
+    }
+    // quantize into the per-wave transpose scratch (accumulator layout:
+    // col = lane&15, row = 4*(lane>>4)+reg — scattered element stores
+    // go to LDS, the global stores below are contiguous 16-B runs)
+#pragma unroll
+    for (int sti = 0; sti < 4; ++sti)
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg)
-        TT::store1(reinterpret_cast<T*>(&sm.cs[wave][4 * kb + reg][0]) +
-                       16 * sti + jcol,
-                   acc[reg]);
-    }
+        TT::storeQ1(reinterpret_cast<T*>(&sm.cs[wave][4 * kb + reg][0]) +
+                        16 * sti + jcol,
+                    accs[sti][reg]);
     {
       const int r2 = lane >> 2;
       const int c0 = (lane & 3) * 16;  // 16 consecutive elements per lane
