@@ -1364,26 +1364,36 @@ __global__ __launch_bounds__(M_NT) void conv_mfma_k(
     }
 
     const int ty0 = tyBase + t * M_TILE_Y;
-    // four 16-element output subtiles per wave, computed in one phase so
-    // the A-fragment ds_reads of subtile i+1 and the MFMA chains of
-    // subtile i interleave (the old per-subtile compute→quantize order
-    // stalled on each accumulator readback right after its 5-chain: the
-    // r01 PMC showed SQ_WAIT_ANY at 48 % of wave cycles on this kernel);
-    // quantization happens in a second phase over all 16 accumulators.
-    f32x4v accs[4];
+    // four 16-element output subtiles per wave, MFMAs issued round-robin
+    // across the four accumulator chains per filter row dy — between two
+    // updates of the SAME accumulator sit three other MFMAs plus the
+    // next row's four A-fragment ds_reads, hiding the MFMA latency that
+    // the chain-per-subtile order exposed (r01/r02 PMC: SQ_WAIT_ANY
+    // 48-55 % on this kernel). A-fragments double-buffer one dy ahead.
+    f32x4v accs[4] = {{0.f, 0.f, 0.f, 0.f},
+                      {0.f, 0.f, 0.f, 0.f},
+                      {0.f, 0.f, 0.f, 0.f},
+                      {0.f, 0.f, 0.f, 0.f}};
+    f16x8v afr0[4], afr1[4];
+    auto loadA = [&](f16x8v (&af)[4], int dy) {
+      const int row = jcol + dy;  // A row i = lane&15, plus filter row
 #pragma unroll
-    for (int sti = 0; sti < 4; ++sti) {
-      const int st = 4 * wave + sti;
-      const int q = 2 * st + 1 + kb;
-      f32x4v acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int dy = 0; dy < K; ++dy) {
-        const int row = jcol + dy;  // A row i = lane&15, plus filter row
-        const f16x8v a = *reinterpret_cast<const f16x8v*>(
+      for (int sti = 0; sti < 4; ++sti) {
+        const int q = 2 * (4 * wave + sti) + 1 + kb;
+        af[sti] = *reinterpret_cast<const f16x8v*>(
             &lds[t & 1][row][(q ^ (row & 7)) * 8]);
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_f16(a, bfrag[dy], acc, 0, 0, 0);
       }
-      accs[sti] = acc;
+    };
+    loadA(afr0, 0);
+#pragma unroll
+    for (int dy = 0; dy < K; ++dy) {
+      f16x8v (&cur)[4] = (dy & 1) ? afr1 : afr0;
+      f16x8v (&nxt)[4] = (dy & 1) ? afr0 : afr1;
+      if (dy + 1 < K) loadA(nxt, dy + 1);
+#pragma unroll
+      for (int sti = 0; sti < 4; ++sti)
+        accs[sti] = __builtin_amdgcn_mfma_f32_16x16x32_f16(
+            cur[sti], bfrag[dy], accs[sti], 0, 0, 0);
     }
     // quantize into the per-wave transpose scratch (accumulator layout:
     // col = lane&15, row = 4*(lane>>4)+reg — scattered element stores
