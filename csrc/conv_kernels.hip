@@ -63,6 +63,15 @@ constexpr int LDS_P4 = WIN_X / 4 + 1;     // LDS row pitch in float4 (+1 pad)
 constexpr int CHUNKS_PER_ROW = WIN_X / 16;
 
 typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef float f32x2 __attribute__((ext_vector_type(2)));
+
+// 2-wide vector fma: forces v_pk_fma_f32 (VOP3P packed math, 2 FMAs per
+// instruction). The scalar form left most taps as v_fmac_f32 — the
+// fused fp16/fp32 kernels are VALU-issue-bound once fusion removes the
+// HBM bound, so packing the tap loop is the direct lever (r02).
+static __device__ __forceinline__ f32x2 fma2(f32x2 a, f32x2 b, f32x2 c) {
+  return __builtin_elementwise_fma(a, b, c);
+}
 // LDS-addrspace volatile vector: volatile forbids LLVM from narrowing or
 // splitting the load; address_space(3) keeps it a ds_read_b128 (a generic
 // volatile pointer lowers to flat_load instead).
@@ -341,36 +350,45 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile_k(
       }
     }
 
-    float acc[2][VEC];
+    f32x2 acc2[2][VEC / 2];
 #pragma unroll
     for (int j = 0; j < 2; ++j)
 #pragma unroll
-      for (int p = 0; p < VEC; ++p) acc[j][p] = 0.f;
+      for (int h = 0; h < VEC / 2; ++h) acc2[j][h] = f32x2{0.f, 0.f};
 
 #pragma unroll
     for (int dy = 0; dy < K; ++dy)
 #pragma unroll
       for (int dx = 0; dx < K; ++dx) {
         const float c = w[dy * K + dx];
+        const f32x2 cc = {c, c};
 #pragma unroll
         for (int j = 0; j < 2; ++j)
 #pragma unroll
-          for (int p = 0; p < VEC; ++p)
-            acc[j][p] =
-                fmaf(c, win[j + dy][4 * NL - SR + p + S * dx], acc[j][p]);
+          for (int h = 0; h < VEC / 2; ++h) {
+            const int o = 4 * NL - SR + 2 * h + S * dx;
+            const f32x2 v = {win[j + dy][o], win[j + dy][o + 1]};
+            acc2[j][h] = fma2(cc, v, acc2[j][h]);
+          }
       }
 
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
       const int y = ty0 + 2 * tyy + j;
       if (y >= py1) continue;
+      float acc[VEC];
+#pragma unroll
+      for (int h = 0; h < VEC / 2; ++h) {
+        acc[2 * h] = acc2[j][h].x;
+        acc[2 * h + 1] = acc2[j][h].y;
+      }
       T* out = dst + (size_t)y * pitchE + ex;
       if (ex >= ex0 && ex + VEC <= ex1) {
-        TT::store4(out, acc[j]);
+        TT::store4(out, acc);
       } else {
 #pragma unroll
         for (int p = 0; p < VEC; ++p)
-          if (ex + p >= ex0 && ex + p < ex1) TT::store1(out + p, acc[j][p]);
+          if (ex + p >= ex0 && ex + p < ex1) TT::store1(out + p, acc[p]);
       }
     }
   };
@@ -1138,23 +1156,26 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile2_k(
             win[rw][4 * b + 3] = v.w;
           }
         }
-        float acc[2][4];
+        f32x2 acc2[2][2];
 #pragma unroll
         for (int j = 0; j < 2; ++j)
 #pragma unroll
-          for (int p = 0; p < 4; ++p) acc[j][p] = 0.f;
+          for (int h = 0; h < 2; ++h) acc2[j][h] = f32x2{0.f, 0.f};
 #pragma unroll
         for (int dy = 0; dy < K; ++dy)
 #pragma unroll
           for (int dx = 0; dx < K; ++dx) {
             const float cw = w[dy * K + dx];
+            const f32x2 cc = {cw, cw};
 #pragma unroll
             for (int j = 0; j < 2; ++j) {
               if (j >= nrows) break;
 #pragma unroll
-              for (int p = 0; p < 4; ++p)
-                acc[j][p] =
-                    fmaf(cw, win[j + dy][4 * NL - SR + p + S * dx], acc[j][p]);
+              for (int h = 0; h < 2; ++h) {
+                const int o = 4 * NL - SR + 2 * h + S * dx;
+                const f32x2 v = {win[j + dy][o], win[j + dy][o + 1]};
+                acc2[j][h] = fma2(cc, v, acc2[j][h]);
+              }
             }
           }
         const int xe = etx0 - 4 * NL + 4 * c;
@@ -1163,11 +1184,13 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile2_k(
           if (j >= nrows) break;
           const int y1 = ty0 - R + r1 + j;
           const bool rowIn = y1 >= m_py0 && y1 < m_py1;
+          const float accj[4] = {acc2[j][0].x, acc2[j][0].y, acc2[j][1].x,
+                                 acc2[j][1].y};
           f32x4 q;
 #pragma unroll
           for (int p = 0; p < 4; ++p) {
             const bool in = rowIn && xe + p >= m_ex0 && xe + p < m_ex1;
-            q[p] = in ? TT::quantF(acc[j][p]) : 0.f;
+            q[p] = in ? TT::quantF(accj[p]) : 0.f;
           }
           sm.mid[r1 + j][c] = q;
         }
@@ -1203,34 +1226,43 @@ __global__ __launch_bounds__(NTHREADS) void conv_tile2_k(
             win[rw][4 * b + 3] = v.w;
           }
         }
-        float acc[2][VEC];
+        f32x2 acc2[2][VEC / 2];
 #pragma unroll
         for (int j = 0; j < 2; ++j)
 #pragma unroll
-          for (int p = 0; p < VEC; ++p) acc[j][p] = 0.f;
+          for (int h = 0; h < VEC / 2; ++h) acc2[j][h] = f32x2{0.f, 0.f};
 #pragma unroll
         for (int dy = 0; dy < K; ++dy)
 #pragma unroll
           for (int dx = 0; dx < K; ++dx) {
             const float cw = w[dy * K + dx];
+            const f32x2 cc = {cw, cw};
 #pragma unroll
             for (int j = 0; j < 2; ++j)
 #pragma unroll
-              for (int p = 0; p < VEC; ++p)
-                acc[j][p] = fmaf(
-                    cw, win[j + dy][4 * NL - SR + p + S * dx], acc[j][p]);
+              for (int h = 0; h < VEC / 2; ++h) {
+                const int o = 4 * NL - SR + 2 * h + S * dx;
+                const f32x2 v = {win[j + dy][o], win[j + dy][o + 1]};
+                acc2[j][h] = fma2(cc, v, acc2[j][h]);
+              }
           }
 #pragma unroll
         for (int j = 0; j < 2; ++j) {
           const int y = ty0 + 2 * tyy + j;
           if (y >= py1) continue;
+          float acc[VEC];
+#pragma unroll
+          for (int h = 0; h < VEC / 2; ++h) {
+            acc[2 * h] = acc2[j][h].x;
+            acc[2 * h + 1] = acc2[j][h].y;
+          }
           T* out = dst + (size_t)y * pitchE + ex;
           if (ex >= ex0 && ex + VEC <= ex1) {
-            TT::store4(out, acc[j]);
+            TT::store4(out, acc);
           } else {
 #pragma unroll
             for (int p = 0; p < VEC; ++p)
-              if (ex + p >= ex0 && ex + p < ex1) TT::store1(out + p, acc[j][p]);
+              if (ex + p >= ex0 && ex + p < ex1) TT::store1(out + p, acc[p]);
           }
         }
       }
