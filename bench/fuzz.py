@@ -23,7 +23,7 @@ from parallel_convolution_amd.utils.rawio import synthetic_image
 
 
 def random_filter(rng):
-    kind = rng.integers(0, 5)
+    kind = rng.integers(0, 6)
     if kind == 0:
         return F.GAUSSIAN_BLUR_3, True
     if kind == 1:
@@ -34,6 +34,17 @@ def random_filter(rng):
         # random dyadic non-negative k=3 (dot-path eligible shapes)
         n = rng.integers(0, 8, (3, 3)).astype(np.float32)
         n[1, 1] += 1
+        shift = 2 ** int(np.ceil(np.log2(max(n.sum(), 1))))
+        return (n / shift).astype(np.float32), True
+    if kind == 4:
+        # random separable non-negative dyadic (sep-path eligible:
+        # sum(a)*sum(b) <= 256 keeps the u16 SWAR bound)
+        k = int(rng.choice([3, 5]))
+        a = rng.integers(0, 4, k)
+        b = rng.integers(0, 4, k)
+        a[k // 2] += 1
+        b[k // 2] += 1
+        n = np.outer(b, a).astype(np.float64)
         shift = 2 ** int(np.ceil(np.log2(max(n.sum(), 1))))
         return (n / shift).astype(np.float32), True
     k = int(rng.choice([3, 5, 7]))
