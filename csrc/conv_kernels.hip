@@ -871,6 +871,143 @@ __global__ __launch_bounds__(NTHREADS, 4) void conv_u8sep_k(
   }
 }
 
+// ---- direct-global separable u8 kernel (LDS-free variant) -----------------
+//
+// Same separable integer math as conv_u8sep_k, but with NO LDS, NO
+// barriers and NO staging pipeline: each lane owns 4 px and sweeps a
+// column chain downward, carrying its 2R boundary h-rows in registers
+// and reading each new input row directly from global (one 12-byte
+// dwordx3 per lane per row — consecutive lanes overlap 8 of those 12
+// bytes, which the vector L1 absorbs; each fresh byte still comes from
+// HBM exactly once, so streaming traffic is unchanged). The win over
+// the LDS version is structural: no staging writes/reads, no workgroup
+// barriers, no prologue re-staging per chain — the whole kernel is the
+// arithmetic. Rotation is a 3-phase (2R+1-phase) unroll so every h
+// index is compile-time. Loads prefetch 3 rows ahead through the
+// rotating preload registers. A/B-selected vs conv_u8sep_k by
+// PCONV_SEP_DIRECT (launch_conv_u8sep).
+struct b12 {
+  unsigned x, y, z;
+};
+
+template <int R>
+__global__ __launch_bounds__(NTHREADS, 8) void conv_u8sepd_k(
+    const unsigned char* __restrict__ src, unsigned char* __restrict__ dst,
+    unsigned wA, unsigned a4, unsigned bvp0, unsigned bvp1, unsigned bvp2,
+    int shift, int pitchE, int bufh, int ex0, int ex1, int py0, int py1,
+    int abase, int tilesX, int chainRows, int nChains) {
+  static_assert(R == 1, "direct-global variant: k=3 (headline) only");
+  constexpr int TXE = 1024;  // 4 px per lane, 256 threads along x
+  const int tid = threadIdx.x;
+  const unsigned bvp[3] = {bvp0, bvp1, bvp2};
+  const unsigned roundC = ((1u << (shift - 1)) - 1u) * 0x00010001u;
+
+  const int nWork = tilesX * nChains;
+  for (int work = blockIdx.x; work < nWork; work += gridDim.x) {
+    const int bx = work % tilesX;
+    const int bc = work / tilesX;
+    const int etx0 = abase + TXE * bx;
+    const int y0c = py0 + chainRows * bc;
+    const int rows = min(chainRows, py1 - y0c);
+    if (rows <= 0) continue;
+
+    // this lane's 12-byte window origin (element etx0 + 4*tid - 4).
+    // Each dword address clamps into the row independently: a clamped
+    // dword only ever feeds outputs that the store masks off (outputs
+    // needing input below element 0 / beyond the pitch cannot be inside
+    // any region the engine launches), and clamping keeps every load
+    // inside the allocation on the first/last rows.
+    const int gxb = etx0 + 4 * tid - 4;
+    const int ax = min(max(gxb, 0), pitchE - 4);
+    const int ay = min(gxb + 4, pitchE - 4);
+    const int az = min(gxb + 8, pitchE - 4);
+
+    auto loadrow = [&](int y, b12& pre) {
+      const unsigned char* p = src + (size_t)min(y, bufh - 1) * pitchE;
+      pre.x = *reinterpret_cast<const unsigned*>(p + ax);
+      pre.y = *reinterpret_cast<const unsigned*>(p + ay);
+      pre.z = *reinterpret_cast<const unsigned*>(p + az);
+    };
+    // horizontal pass from a preloaded row: h pair dwords for 4 px
+    auto hrow = [&](const b12& d, unsigned* hp) {
+      unsigned h[4];
+#pragma unroll
+      for (int p = 0; p < 4; ++p) {
+        const unsigned pk =
+            (p < 3) ? __builtin_amdgcn_perm(
+                          d.y, d.x,
+                          (unsigned)(3 + p) | ((unsigned)(4 + p) << 8) |
+                              ((unsigned)(5 + p) << 16))
+                    : __builtin_amdgcn_perm(d.z, d.y, 2u | (3u << 8) | (4u << 16));
+        h[p] = __builtin_amdgcn_udot4(pk, wA, 0u, false);
+      }
+      hp[0] = __builtin_amdgcn_perm(h[1], h[0], 0x05040100u);
+      hp[1] = __builtin_amdgcn_perm(h[3], h[2], 0x05040100u);
+    };
+    const int ex = etx0 + VEC * tid;
+    auto vstore = [&](int y, const unsigned (&h0)[2], const unsigned (&h1)[2],
+                      const unsigned (&h2)[2]) {
+      if (y >= py1) return;
+      unsigned acc0 = pk_mad_u16(bvp[0], h0[0], roundC);
+      unsigned acc1 = pk_mad_u16(bvp[0], h0[1], roundC);
+      acc0 = pk_mad_u16(bvp[1], h1[0], acc0);
+      acc1 = pk_mad_u16(bvp[1], h1[1], acc1);
+      acc0 = pk_mad_u16(bvp[2], h2[0], acc0);
+      acc1 = pk_mad_u16(bvp[2], h2[1], acc1);
+      acc0 += (acc0 >> shift) & 0x00010001u;
+      acc1 += (acc1 >> shift) & 0x00010001u;
+      acc0 >>= shift;
+      acc1 >>= shift;
+      const unsigned qb = __builtin_amdgcn_perm(acc1, acc0, 0x06040200u);
+      unsigned char* out = dst + (size_t)y * pitchE + ex;
+      if (ex >= ex0 && ex + VEC <= ex1) {
+        *reinterpret_cast<unsigned*>(out) = qb;
+      } else {
+#pragma unroll
+        for (int p = 0; p < VEC; ++p)
+          if (ex + p >= ex0 && ex + p < ex1)
+            out[p] = (unsigned char)((qb >> (8 * p)) & 0xffu);
+      }
+    };
+
+    // 3-phase software pipeline: pre[i]/h-ring rotate with period 3, so
+    // all indices stay compile-time. Prefetch depth 3 rows.
+    b12 pA, pB, pC;
+    unsigned h0[2], h1[2], h2[2];
+    loadrow(y0c - 1, pA);
+    loadrow(y0c, pB);
+    loadrow(y0c + 1, pC);
+    hrow(pA, h0);
+    loadrow(y0c + 2, pA);  // a consumed preload re-issues 3 rows ahead
+    hrow(pB, h1);
+    loadrow(y0c + 3, pB);
+    // steady state, 3-phase: each phase consumes one preload into the h
+    // ring, immediately re-issues that register 3 rows ahead (full
+    // triple of prefetch distance), and emits one output row
+    int j = 0;
+    for (; j + 3 <= rows; j += 3) {
+      hrow(pC, h2);
+      loadrow(y0c + j + 4, pC);
+      vstore(y0c + j, h0, h1, h2);
+      hrow(pA, h0);
+      loadrow(y0c + j + 5, pA);
+      vstore(y0c + j + 1, h1, h2, h0);
+      hrow(pB, h1);
+      loadrow(y0c + j + 6, pB);
+      vstore(y0c + j + 2, h2, h0, h1);
+    }
+    // remainder (<=2 rows)
+    if (j < rows) {
+      hrow(pC, h2);
+      vstore(y0c + j, h0, h1, h2);
+      if (j + 1 < rows) {
+        hrow(pA, h0);
+        vstore(y0c + j + 1, h1, h2, h0);
+      }
+    }
+  }
+}
+
 // ---- fused integer-dot u8 double-iteration kernel -------------------------
 //
 // conv_u8dot_k x conv_tile2_k: two exact-integer iterations per launch
@@ -1795,6 +1932,30 @@ void launch_conv_u8sep(const void* src, void* dst, const int* sep, int k,
                        const BlockGeom& g, const Region& r,
                        hipStream_t stream) {
   if (r.py1 <= r.py0 || r.ex1 <= r.ex0) return;
+  // LDS-free direct-global variant (A/B, k=3 only): PCONV_SEP_DIRECT=1
+  static const int direct = [] {
+    const char* e = getenv("PCONV_SEP_DIRECT");
+    return e ? atoi(e) : 0;
+  }();
+  if (k == 3 && direct) {
+    const int TXE = 1024;
+    const int abase = (r.ex0 / 16) * 16;
+    const int tilesX = (r.ex1 - abase + TXE - 1) / TXE;
+    const int rows = r.py1 - r.py0;
+    const int chainRows = std::max(24, std::min(96, (rows / 48) * 3));
+    const int chains = (rows + chainRows - 1) / chainRows;
+    const int grid =
+        grid_for((const void*)&conv_u8sepd_k<1>, tilesX * chains);
+    const unsigned* a = reinterpret_cast<const unsigned*>(sep + 1);
+    const unsigned* b = reinterpret_cast<const unsigned*>(sep + 6);
+    const unsigned wA = a[0] | (a[1] << 8) | (a[2] << 16);
+    hipLaunchKernelGGL((conv_u8sepd_k<1>), dim3(grid), dim3(NTHREADS), 0,
+                       stream, (const unsigned char*)src, (unsigned char*)dst,
+                       wA, 0u, b[0] * 0x10001u, b[1] * 0x10001u,
+                       b[2] * 0x10001u, sep[0], g.pitchE, g.bufh, r.ex0,
+                       r.ex1, r.py0, r.py1, abase, tilesX, chainRows, chains);
+    return;
+  }
   if (k == 3)
     launch_u8sep_r<1>(src, dst, sep, g, r, stream);
   else
