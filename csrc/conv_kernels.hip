@@ -1879,7 +1879,10 @@ void launch_u8dot_impl(const void* src, void* dst, const unsigned* w9,
 // per resident workgroup (grid cap is ~1024 at 4 WG/CU).
 inline int chain_for_wide(int tilesX, int tilesY) {
   if (chain_override() > 0) return chain_override();
-  return std::max(1, std::min(16, (tilesX * tilesY) / 1024));
+  // cap 12: measured best on the headline sweep (PCONV_CHAIN 8/12/16/24
+  // -> 1982/2061/1991/1932) — long chains serialize the tail of each
+  // grid-stride pass
+  return std::max(1, std::min(12, (tilesX * tilesY) / 1024));
 }
 
 template <int R>
