@@ -219,18 +219,36 @@ def test_mfma_engine_selected_and_byte_exact():
     spec = ProblemSpec(512, 384, 1, DType.U8, GAUSSIAN_BLUR_3, 10)
     img = synthetic_image(384, 512, 1, DType.U8)
     ref = run_reference(spec, img)
+    # the integer sep/dot paths outrank MFMA in launch_region — disable
+    # both so FORCE_MFMA really exercises conv_mfma_k (without this the
+    # test silently reran the integer kernel)
     os.environ["PCONV_FORCE_MFMA"] = "1"
+    os.environ["PCONV_DISABLE_SEP"] = "1"
+    os.environ["PCONV_DISABLE_DOT"] = "1"
     try:
         assert mfma_eligible(spec)
-        out_mfma = _engine_mod().run_single_gpu(spec, img)
+        eng_mod = _engine_mod()
+        from parallel_convolution_amd.parallel.decomp import make_blocks
+
+        probe = eng_mod.HipEngine(spec, make_blocks(384, 512, 1, 1))
+        assert probe.kernel_path() == "mfma-f16"
+        del probe
+        out_mfma = eng_mod.run_single_gpu(spec, img)
     finally:
         del os.environ["PCONV_FORCE_MFMA"]
+        del os.environ["PCONV_DISABLE_SEP"]
+        del os.environ["PCONV_DISABLE_DOT"]
     np.testing.assert_array_equal(out_mfma, ref)
     os.environ["PCONV_DISABLE_MFMA"] = "1"
+    os.environ["PCONV_DISABLE_SEP"] = "1"
+    os.environ["PCONV_DISABLE_DOT"] = "1"
+    os.environ["PCONV_FUSE"] = "0"
     try:
         out_valu = _engine_mod().run_single_gpu(spec, img)
     finally:
-        del os.environ["PCONV_DISABLE_MFMA"]
+        for v in ("PCONV_DISABLE_MFMA", "PCONV_DISABLE_SEP",
+                  "PCONV_DISABLE_DOT", "PCONV_FUSE"):
+            del os.environ[v]
     np.testing.assert_array_equal(out_valu, out_mfma)
 
 
@@ -248,10 +266,13 @@ def test_mfma_variants_match_oracle(W, H, C, dt, filt, iters):
     spec = ProblemSpec(W, H, C, dt, filt, iters)
     img = synthetic_image(H, W, C, dt)
     os.environ["PCONV_FORCE_MFMA"] = "1"
+    os.environ["PCONV_DISABLE_SEP"] = "1"  # sep/dot outrank MFMA
+    os.environ["PCONV_DISABLE_DOT"] = "1"
     try:
         out = _engine_mod().run_single_gpu(spec, img)
     finally:
-        del os.environ["PCONV_FORCE_MFMA"]
+        for v in ("PCONV_FORCE_MFMA", "PCONV_DISABLE_SEP", "PCONV_DISABLE_DOT"):
+            del os.environ[v]
     ref = run_reference(spec, img)
     assert_matches(spec, out, ref)
 
@@ -262,10 +283,13 @@ def test_mfma_multiblock_protocol():
     spec = ProblemSpec(96, 80, 1, DType.U8, GAUSSIAN_BLUR_3, 20)
     img = synthetic_image(80, 96, 1, DType.U8)
     os.environ["PCONV_FORCE_MFMA"] = "1"
+    os.environ["PCONV_DISABLE_SEP"] = "1"  # sep/dot outrank MFMA
+    os.environ["PCONV_DISABLE_DOT"] = "1"
     try:
         out = _engine_mod().run_local_blocks_gpu(spec, img, 8)
     finally:
-        del os.environ["PCONV_FORCE_MFMA"]
+        for v in ("PCONV_FORCE_MFMA", "PCONV_DISABLE_SEP", "PCONV_DISABLE_DOT"):
+            del os.environ[v]
     ref = run_reference(spec, img)
     np.testing.assert_array_equal(out, ref)
 
@@ -435,11 +459,16 @@ def test_fused_dot_byte_exact(p):
 
     spec = ProblemSpec(96, 80, 1, DType.U8, GAUSSIAN_BLUR_3, 21)
     img = synthetic_image(80, 96, 1, DType.U8)
+    # PCONV_DISABLE_SEP: since r02 the separable path outranks the dot
+    # path for blur3 and disables dot fusion — without this the test
+    # would silently exercise conv_u8sep_k instead of conv_u8dot2_k
     os.environ["PCONV_DOT_FUSE"] = "1"
+    os.environ["PCONV_DISABLE_SEP"] = "1"
     try:
         out = _engine_mod().run_local_blocks_gpu(spec, img, p)
     finally:
         del os.environ["PCONV_DOT_FUSE"]
+        del os.environ["PCONV_DISABLE_SEP"]
     ref = run_reference(spec, img)
     np.testing.assert_array_equal(out, ref)
 
