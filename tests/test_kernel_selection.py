@@ -156,3 +156,51 @@ def test_sep_rounding_edge_values():
     q = (v + t) >> shift
     ref = np.rint(s.astype(np.float64) / (1 << shift)).astype(np.uint32)
     np.testing.assert_array_equal(q, ref)
+
+
+def test_sep_factorization_property():
+    """500 random cases: whenever u8sep_weights accepts, the returned
+    integer factorization must reconstruct the filter EXACTLY and satisfy
+    every kernel precondition (u16 SWAR bound, byte-sized horizontal
+    taps); random separable non-negative dyadic filters within bounds
+    must always be accepted."""
+    rng = np.random.default_rng(42)
+    accepted = 0
+    for trial in range(500):
+        k = int(rng.choice([3, 5]))
+        mode = rng.integers(0, 3)
+        if mode == 0:
+            # separable dyadic within bounds
+            a = rng.integers(0, 4, k)
+            b = rng.integers(0, 4, k)
+            a[k // 2] += 1
+            b[k // 2] += 1
+            n = np.outer(b, a).astype(np.float64)
+            shift = 2 ** int(np.ceil(np.log2(max(n.sum(), 1))))
+            f = (n / shift).astype(np.float32)
+            must_accept = 255 * int(a.sum()) * int(b.sum()) + shift // 2 <= 65535
+        elif mode == 1:
+            # random dyadic, usually rank>1
+            n = rng.integers(0, 5, (k, k)).astype(np.float64)
+            n[k // 2, k // 2] += 1
+            f = (n / 2 ** int(np.ceil(np.log2(max(n.sum(), 1))))).astype(np.float32)
+            must_accept = None
+        else:
+            f = rng.random((k, k), dtype=np.float32)
+            f /= f.sum()
+            must_accept = None
+        s = ProblemSpec(32, 32, 1, DType.U8, f, 1)
+        w = u8sep_weights(s)
+        if must_accept and not w:
+            raise AssertionError(f"trial {trial}: in-bounds separable rejected: {f}")
+        if w:
+            accepted += 1
+            shift_, a_, b_ = w[0], np.array(w[1:6][:k]), np.array(w[6:11][:k])
+            assert (a_ >= 0).all() and (b_ >= 0).all()
+            assert a_.max() <= 255
+            assert 255 * int(a_.sum()) * int(b_.sum()) + (1 << (shift_ - 1)) <= 65535
+            np.testing.assert_array_equal(
+                np.outer(b_, a_) / np.float64(1 << shift_),
+                np.asarray(f, dtype=np.float64),
+            )
+    assert accepted >= 100  # the separable class must actually be exercised
