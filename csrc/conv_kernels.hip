@@ -1445,6 +1445,10 @@ constexpr int M_QPR = 36;  // staged 16-B quarters per row (288-element window)
 constexpr int M_NT = 256;  // MFMA kernel workgroup (4 waves of 16x16 subtile quads)
 
 template <typename TT, int R, int S>
+// NOTE: no min-blocks hint here — forcing 5 WG/CU (<=102 VGPRs) makes
+// the R=2 u8 variant (the RGB-5×5 production kernel) spill 96 B of
+// scratch; the default allocation (110 VGPRs, 4 waves/SIMD) measures
+// faster than the squeezed+spilling 5-wave build.
 __global__ __launch_bounds__(M_NT) void conv_mfma_k(
     const typename TT::T* __restrict__ src, typename TT::T* __restrict__ dst,
     const float* __restrict__ wgt, int pitchE, int bufh, int ex0, int ex1,
