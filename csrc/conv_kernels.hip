@@ -653,12 +653,14 @@ static __device__ __forceinline__ unsigned pk_mad_u16(unsigned a, unsigned b,
   return d;
 }
 
-// second launch_bounds arg: 4 workgroups/CU is the LDS-bound occupancy
-// ceiling (34.3 KB of 160 KB each), so give the allocator the full
-// 512/(4 waves) = 128-VGPR budget — the default heuristic picked 97 and
-// spilled the prefetch registers to scratch, defeating the pipeline.
-template <int R>
-__global__ __launch_bounds__(NTHREADS, 4) void conv_u8sep_k(
+// RING/MINB are template parameters for same-box A/B (box-to-box
+// variance exceeds the effect size): RING=30 rows = 31.7 KB LDS → 5
+// WG/CU (5-wave VGPR budget 102 holds the kernel's natural ~97 without
+// spills); RING=32 pow2 = 33.8 KB → 4 WG/CU with a cheaper slot AND.
+// Without a min-blocks hint the default heuristic spilled the prefetch
+// registers to scratch, defeating the pipeline.
+template <int R, int RINGP = 30, int MINB = 5>
+__global__ __launch_bounds__(NTHREADS, MINB) void conv_u8sep_k(
     const unsigned char* __restrict__ src, unsigned char* __restrict__ dst,
     unsigned wA, unsigned a4, unsigned bvp0, unsigned bvp1, unsigned bvp2,
     unsigned bvp3, unsigned bvp4, int shift, int pitchE, int bufh, int ex0,
@@ -667,9 +669,12 @@ __global__ __launch_bounds__(NTHREADS, 4) void conv_u8sep_k(
   constexpr int K = 2 * R + 1;
   constexpr int TXE = 1024;  // output elements per tile (4 per lane)
   constexpr int ROWS_IN = TILE_Y + 2 * R;
-  constexpr int RING = 32;  // > ROWS_IN + 2*TILE_Y in flight; pow2 slot
+  // LDS rows in flight = current tile's ROWS_IN + next tile's TILE_Y
+  // staged rows (tile t+2 prefetches live in REGISTERS) = 18/20 — any
+  // RINGP >= 20 is correct.
+  constexpr int RING = RINGP;
   constexpr int WINB = TXE + 32;
-  constexpr int PITCHB = WINB + 16;
+  constexpr int PITCHB = WINB;  // rows are 66 16-B quarters exactly
   constexpr int QPR = WINB / 16;
   constexpr int SITEMS = TILE_Y * QPR;
   constexpr int PITEMS = ROWS_IN * QPR;
@@ -692,8 +697,12 @@ __global__ __launch_bounds__(NTHREADS, 4) void conv_u8sep_k(
     const int tyBase = py0 + TILE_Y * chainLen * bc;
     const int nt = min(chainLen, (py1 - tyBase + TILE_Y - 1) / TILE_Y);
 
-    // ring slot of chain-relative input row rr (absolute tyBase - R + rr)
-    auto slot = [&](int rr) { return rr & (RING - 1); };
+    // ring slot of chain-relative input row rr (absolute tyBase - R + rr);
+    // any RING consecutive rr map to distinct slots (span is 18/20);
+    // pow2 RING lowers to an AND, otherwise constant-modulo mul-hi
+    auto slot = [&](int rr) {
+      return (RING & (RING - 1)) == 0 ? (rr & (RING - 1)) : rr % RING;
+    };
 
     int srow[SNITER], sq[SNITER], ga[SNITER];
 #pragma unroll
@@ -1889,7 +1898,7 @@ inline int chain_for_wide(int tilesX, int tilesY) {
   return std::max(1, std::min(12, (tilesX * tilesY) / 1024));
 }
 
-template <int R>
+template <int R, int RINGP = 30, int MINB = 5>
 void launch_u8sep_r(const void* src, void* dst, const int* sep,
                     const BlockGeom& g, const Region& r, hipStream_t stream) {
   // sep = [shift, a0..a4, b0..b4] (hip_engine.u8sep_weights layout)
@@ -1899,7 +1908,8 @@ void launch_u8sep_r(const void* src, void* dst, const int* sep,
   const int tilesY = (r.py1 - r.py0 + TILE_Y - 1) / TILE_Y;
   const int chainLen = chain_for_wide(tilesX, tilesY);
   const int chains = (tilesY + chainLen - 1) / chainLen;
-  const int grid = grid_for((const void*)&conv_u8sep_k<R>, tilesX * chains);
+  const int grid =
+      grid_for((const void*)&conv_u8sep_k<R, RINGP, MINB>, tilesX * chains);
   const unsigned* a = reinterpret_cast<const unsigned*>(sep + 1);
   const unsigned* b = reinterpret_cast<const unsigned*>(sep + 6);
   const unsigned wA = a[0] | (a[1] << 8) | (a[2] << 16) |
@@ -1907,11 +1917,11 @@ void launch_u8sep_r(const void* src, void* dst, const int* sep,
   const unsigned a4 = R == 2 ? a[4] : 0u;
   unsigned bvp[5] = {};
   for (int j = 0; j < 2 * R + 1; ++j) bvp[j] = b[j] * 0x00010001u;
-  hipLaunchKernelGGL((conv_u8sep_k<R>), dim3(grid), dim3(NTHREADS), 0, stream,
-                     (const unsigned char*)src, (unsigned char*)dst, wA, a4,
-                     bvp[0], bvp[1], bvp[2], bvp[3], bvp[4], sep[0], g.pitchE,
-                     g.bufh, r.ex0, r.ex1, r.py0, r.py1, abase, tilesX,
-                     chainLen, chains);
+  hipLaunchKernelGGL((conv_u8sep_k<R, RINGP, MINB>), dim3(grid),
+                     dim3(NTHREADS), 0, stream, (const unsigned char*)src,
+                     (unsigned char*)dst, wA, a4, bvp[0], bvp[1], bvp[2],
+                     bvp[3], bvp[4], sep[0], g.pitchE, g.bufh, r.ex0, r.ex1,
+                     r.py0, r.py1, abase, tilesX, chainLen, chains);
 }
 
 void launch_u8dot2_impl(const void* src, void* dst, const unsigned* w9,
@@ -1961,6 +1971,19 @@ void launch_conv_u8sep(const void* src, void* dst, const int* sep, int k,
                        wA, 0u, b[0] * 0x10001u, b[1] * 0x10001u,
                        b[2] * 0x10001u, sep[0], g.pitchE, g.bufh, r.ex0,
                        r.ex1, r.py0, r.py1, abase, tilesX, chainRows, chains);
+    return;
+  }
+  // ring-size A/B (same-box measurement; box variance exceeds the
+  // effect): PCONV_SEP_RING=32 selects the pow2 32-ring / 4 WG variant
+  static const int ring32 = [] {
+    const char* e = getenv("PCONV_SEP_RING");
+    return e && atoi(e) == 32;
+  }();
+  if (ring32) {
+    if (k == 3)
+      launch_u8sep_r<1, 32, 4>(src, dst, sep, g, r, stream);
+    else
+      launch_u8sep_r<2, 32, 4>(src, dst, sep, g, r, stream);
     return;
   }
   if (k == 3)
