@@ -1973,23 +1973,26 @@ void launch_conv_u8sep(const void* src, void* dst, const int* sep, int k,
                        r.ex1, r.py0, r.py1, abase, tilesX, chainRows, chains);
     return;
   }
-  // ring-size A/B (same-box measurement; box variance exceeds the
-  // effect): PCONV_SEP_RING=32 selects the pow2 32-ring / 4 WG variant
-  static const int ring32 = [] {
+  // Ring-size/occupancy A/B, measured interleaved on ONE box
+  // (profiles/r02): 32-ring/4WG 2011-2037 vs 30-ring/5WG 1948-1987 at
+  // steps=100 — the kernel is issue-bound, extra waves don't pay and
+  // the modulo slot + tighter VGPR allocation cost ~3 %. Default 32;
+  // PCONV_SEP_RING=30 keeps the 5-wave variant measurable.
+  static const int ring30 = [] {
     const char* e = getenv("PCONV_SEP_RING");
-    return e && atoi(e) == 32;
+    return e && atoi(e) == 30;
   }();
-  if (ring32) {
+  if (ring30) {
     if (k == 3)
-      launch_u8sep_r<1, 32, 4>(src, dst, sep, g, r, stream);
+      launch_u8sep_r<1, 30, 5>(src, dst, sep, g, r, stream);
     else
-      launch_u8sep_r<2, 32, 4>(src, dst, sep, g, r, stream);
+      launch_u8sep_r<2, 30, 5>(src, dst, sep, g, r, stream);
     return;
   }
   if (k == 3)
-    launch_u8sep_r<1>(src, dst, sep, g, r, stream);
+    launch_u8sep_r<1, 32, 4>(src, dst, sep, g, r, stream);
   else
-    launch_u8sep_r<2>(src, dst, sep, g, r, stream);
+    launch_u8sep_r<2, 32, 4>(src, dst, sep, g, r, stream);
 }
 
 void launch_conv_u8dot2(const void* src, void* dst, const unsigned* w9,
