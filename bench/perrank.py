@@ -72,6 +72,9 @@ def time_local_grid(n, p, iters, chunk=None):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--iters", type=int, default=300)
+    ap.add_argument("--all-shapes", action="store_true",
+                    help="also time the N=2/N=4 rank shapes and the fp16 "
+                    "weak-scaling rank shape")
     args = ap.parse_args()
 
     # 1. per-rank block of the 2x4 headline decomposition
@@ -84,6 +87,31 @@ def main():
     # 3. LOCAL 2x4 grid on one GPU (D2D exchange, real chunk schedule)
     for chunk in (16, 32):
         time_local_grid(16384, 8, max(50, args.iters // 4), chunk=chunk)
+    if args.all_shapes:
+        # N=2 (1x2 grid) and N=4 (2x2) headline rank shapes
+        time_single(16384, 8192, max(50, args.iters // 2), label="per-rank(1x2, N=2)")
+        time_single(8192, 8192, max(50, args.iters // 2), label="per-rank(2x2, N=4)")
+        # config-5 weak-scaling rank shape: 65536^2 fp16 on 2x4
+        from parallel_convolution_amd.spec import DType as _DT
+
+        spec = ProblemSpec(16384, 32768, 1, _DT.F16, GAUSSIAN_BLUR_3, 1)
+        blocks = make_blocks(32768, 16384, 1, 1, halo=halo_for_radius(1))
+        eng = HipEngine(spec, blocks)
+        eng.load_block(0, synthetic_image(32768, 16384, 1, _DT.F16, seed=5))
+        eng.run(10)
+        torch.cuda.synchronize()
+        import time as _t
+
+        t0 = _t.perf_counter()
+        eng.run(40)
+        torch.cuda.synchronize()
+        dt = _t.perf_counter() - t0
+        print(
+            f"[perrank] fp16-rank(2x4 of 65536^2) 32768x16384 "
+            f"{dt / 40 * 1e6:8.1f} us/iter  {32768 * 16384 * 40 / dt / 1e9:6.0f} Gpx/s "
+            f"kernel={eng.kernel_path()}",
+            flush=True,
+        )
 
 
 if __name__ == "__main__":
