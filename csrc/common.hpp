@@ -88,6 +88,12 @@ bool launch_conv2(DType dt, int S, int k, const void* src, void* dst,
                   const float* wgt, const BlockGeom& g, const Region& rm,
                   const Region& ro, hipStream_t stream);
 
+// Strided<->contiguous 2D device copy (halo pack/unpack):
+// hipMemcpy2DAsync's SDMA path measured ~8x slower than a plain copy
+// kernel for the narrow tall E/W strips (bench/exchange_cost.py).
+void launch_copy2d(void* dst, size_t dpitch, const void* src, size_t spitch,
+                   size_t widthB, size_t rows, hipStream_t stream);
+
 // one v_mfma_f32_16x16x32_f16 with the conv kernel's assumed fragment
 // layout: D[16][16] = A[16][32] . B[32][16] (layout verification probe)
 void launch_mfma_probe(const float* A32, const float* B32, float* D,

@@ -1669,6 +1669,41 @@ __global__ __launch_bounds__(NTHREADS) void conv_generic_k(
   }
 }
 
+// ---- halo pack/unpack ---------------------------------
+//
+// Strided<->contiguous 2D byte copy for the E/W halo strips. Replaces
+// hipMemcpy2DAsync, whose SDMA path measured ~41 µs for a 32 B × 16384
+// row strip (bench/exchange_cost.py) — ~8x the kernel cost of actually
+// moving the 0.5 MB; per 32-iteration chunk that tax is 4 copies and it
+// sits on the exchange critical path the interior overlap must hide.
+// Vectorized 16-B path when geometry allows, byte path otherwise
+// (halo-deepened tail chunks produce arbitrary widths/alignments).
+
+__global__ __launch_bounds__(NTHREADS) void copy2d_b_k(
+    unsigned char* __restrict__ dst, size_t dpitch,
+    const unsigned char* __restrict__ src, size_t spitch, int widthB,
+    long total) {
+  for (long i = (long)blockIdx.x * NTHREADS + threadIdx.x; i < total;
+       i += (long)gridDim.x * NTHREADS) {
+    const long r = i / widthB;
+    const int c = (int)(i - r * widthB);
+    dst[(size_t)r * dpitch + c] = src[(size_t)r * spitch + c];
+  }
+}
+
+__global__ __launch_bounds__(NTHREADS) void copy2d_v_k(
+    unsigned char* __restrict__ dst, size_t dpitch,
+    const unsigned char* __restrict__ src, size_t spitch, int widthV,
+    long totalV) {
+  for (long i = (long)blockIdx.x * NTHREADS + threadIdx.x; i < totalV;
+       i += (long)gridDim.x * NTHREADS) {
+    const long r = i / widthV;
+    const int c = (int)(i - r * widthV) * 16;
+    *reinterpret_cast<uint4*>(dst + (size_t)r * dpitch + c) =
+        *reinterpret_cast<const uint4*>(src + (size_t)r * spitch + c);
+  }
+}
+
 // ---- diff-max (convergence check) ---------------------
 
 __device__ __forceinline__ void atomicMaxFloatPos(float* addr, float v) {
@@ -2036,6 +2071,30 @@ bool launch_conv2(DType dt, int S, int k, const void* src, void* dst,
       return launch2_dtype<F32T>(S, k, src, dst, wgt, g, rm, ro, stream);
   }
   return false;
+}
+
+void launch_copy2d(void* dst, size_t dpitch, const void* src, size_t spitch,
+                   size_t widthB, size_t rows, hipStream_t stream) {
+  if (widthB == 0 || rows == 0) return;
+  const bool vec =
+      widthB % 16 == 0 && dpitch % 16 == 0 && spitch % 16 == 0 &&
+      (reinterpret_cast<uintptr_t>(dst) % 16 == 0) &&
+      (reinterpret_cast<uintptr_t>(src) % 16 == 0);
+  if (vec) {
+    const long totalV = (long)rows * (widthB / 16);
+    const int grid = (int)std::max<long>(
+        1, std::min<long>((totalV + NTHREADS - 1) / NTHREADS, 2048));
+    hipLaunchKernelGGL(copy2d_v_k, dim3(grid), dim3(NTHREADS), 0, stream,
+                       (unsigned char*)dst, dpitch, (const unsigned char*)src,
+                       spitch, (int)(widthB / 16), totalV);
+  } else {
+    const long total = (long)rows * widthB;
+    const int grid = (int)std::max<long>(
+        1, std::min<long>((total + NTHREADS - 1) / NTHREADS, 2048));
+    hipLaunchKernelGGL(copy2d_b_k, dim3(grid), dim3(NTHREADS), 0, stream,
+                       (unsigned char*)dst, dpitch, (const unsigned char*)src,
+                       spitch, (int)widthB, total);
+  }
 }
 
 void launch_mfma_probe(const float* A32, const float* B32, float* D,

@@ -344,8 +344,21 @@ class ConvEngine {
 
   void copy2d(void* dst, size_t dpitch, const void* src, size_t spitch,
               size_t widthB, size_t rows) {
-    HIP_CHECK(hipMemcpy2DAsync(dst, dpitch, src, spitch, widthB, rows,
-                               hipMemcpyDeviceToDevice, comm_));
+    // pack kernel instead of hipMemcpy2DAsync: the SDMA strided path
+    // costs ~41 us for a 32 B x 16384-row strip vs ~5 us kernel
+    // (bench/exchange_cost.py); this sits on the exchange critical path
+    // every chunk. PCONV_SDMA_PACK=1 restores the SDMA path for A/B.
+    static const bool sdma = [] {
+      const char* e = getenv("PCONV_SDMA_PACK");
+      return e && atoi(e) != 0;
+    }();
+    if (sdma) {
+      HIP_CHECK(hipMemcpy2DAsync(dst, dpitch, src, spitch, widthB, rows,
+                                 hipMemcpyDeviceToDevice, comm_));
+    } else {
+      launch_copy2d(dst, dpitch, src, spitch, widthB, rows, comm_);
+      HIP_CHECK(hipGetLastError());
+    }
   }
 
   // LOCAL mode: neighbors are local block indices; D2D copies.
