@@ -61,6 +61,15 @@ void launch_conv(DType dt, int S, int k, const void* src, void* dst,
                  const float* wgt, const BlockGeom& g, const Region& r,
                  bool useMfma, hipStream_t stream);
 
+// Element-indexed strip launch (conv_generic_k) for thin border/band
+// regions where the 1024-element tiled kernels waste ~30x on masked
+// lanes. Bit-compatible with every fast path on their eligible inputs:
+// fp32 fma in row-major tap order with the same store rounding — for
+// the dyadic u8 class all sums are fp32-exact, so integer RNE == this.
+void launch_conv_strip(DType dt, int S, int k, const void* src, void* dst,
+                       const float* wgt, const BlockGeom& g, const Region& r,
+                       hipStream_t stream);
+
 // Integer-dot u8 fast path (k=3, S=1, non-negative dyadic filter
 // w = n/2^shift, n in [0,255], sum(n) <= 2^shift): exact vs the fp32
 // contract. w9 = the nine integer taps row-major.

@@ -177,7 +177,15 @@ struct F16T {
     u.y = *reinterpret_cast<unsigned*>(&h1);
     *reinterpret_cast<uint2*>(p) = u;
   }
-  static __device__ __forceinline__ void store1(T* p, float v) { *p = __float2half_rn(v); }
+  static __device__ __forceinline__ void store1(T* p, float v) {
+    // packed-instruction rounding, NOT the scalar v_cvt_f16_f32: the
+    // two break exact ties differently on gfx950 (see quantF) — every
+    // f16 store path must round identically or mixed-kernel runs
+    // (tile interior + generic strips) lose bit-reproducibility.
+    unsigned pk;
+    asm("v_cvt_pk_f16_f32 %0, %1, %1" : "=v"(pk) : "v"(v));
+    *reinterpret_cast<unsigned short*>(p) = (unsigned short)pk;
+  }
   static __device__ __forceinline__ float toF(T v) { return __half2float(v); }
   static __device__ __forceinline__ float quantF(float v) {
     // MUST round exactly like store4/store8 (v_cvt_pk_f16_f32): the
@@ -2055,6 +2063,34 @@ void launch_conv(DType dt, int S, int k, const void* src, void* dst,
       return launch_dtype<F16T>(S, k, src, dst, wgt, g, r, useMfma, stream);
     case DType::F32:
       return launch_dtype<F32T>(S, k, src, dst, wgt, g, r, false, stream);
+  }
+}
+
+void launch_conv_strip(DType dt, int S, int k, const void* src, void* dst,
+                       const float* wgt, const BlockGeom& g, const Region& r,
+                       hipStream_t stream) {
+  if (r.py1 <= r.py0 || r.ex1 <= r.ex0) return;
+  const int R = k / 2;
+  const long total = (long)(r.py1 - r.py0) * (r.ex1 - r.ex0);
+  int blocks = (int)std::min<long>((total + NTHREADS - 1) / NTHREADS, 2048);
+  blocks = std::max(blocks, 1);
+  switch (dt) {
+    case DType::U8:
+      hipLaunchKernelGGL((conv_generic_k<U8T>), dim3(blocks), dim3(NTHREADS),
+                         0, stream, (const unsigned char*)src,
+                         (unsigned char*)dst, wgt, g.pitchE, S, R, r.ex0,
+                         r.ex1, r.py0, r.py1);
+      break;
+    case DType::F16:
+      hipLaunchKernelGGL((conv_generic_k<F16T>), dim3(blocks), dim3(NTHREADS),
+                         0, stream, (const __half*)src, (__half*)dst, wgt,
+                         g.pitchE, S, R, r.ex0, r.ex1, r.py0, r.py1);
+      break;
+    case DType::F32:
+      hipLaunchKernelGGL((conv_generic_k<F32T>), dim3(blocks), dim3(NTHREADS),
+                         0, stream, (const float*)src, (float*)dst, wgt,
+                         g.pitchE, S, R, r.ex0, r.ex1, r.py0, r.py1);
+      break;
   }
 }
 

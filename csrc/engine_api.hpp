@@ -29,6 +29,7 @@
 #include <rccl/rccl.h>
 
 #include <algorithm>
+#include <utility>
 #include <cstdlib>
 #include <cstring>
 
@@ -92,6 +93,15 @@ class ConvEngine {
     } else {
       fusePairs_ = dt_ != DType::U8;
     }
+    // Exchange-overlap depth: how many iterations' INTERIORS launch on
+    // the compute stream while the chunk's halo exchange is in flight
+    // on the comm stream (their border bands replay after the exchange
+    // event). Depth 1 is the classic interior/border split; deeper
+    // overlap covers an exchange longer than one interior iteration
+    // (~19 us at the 8-GPU rank shape) at the cost of 4 extra thin band
+    // launches per depth step. Default 2; PCONV_OVERLAP_DEPTH overrides.
+    const char* oe = getenv("PCONV_OVERLAP_DEPTH");
+    overlapDepth_ = oe ? std::max(1, atoi(oe)) : 2;
   }
 
   ~ConvEngine() {
@@ -167,38 +177,58 @@ class ConvEngine {
       int t0 = 0;
       if (r > 0 && d > 0 && any_neighbor()) {
         // Overlap (BASELINE.json north-star: halo exchange "overlapped
-        // with interior-tile compute on a second HIP stream"): iteration
-        // 0's INTERIOR — the owned block shrunk by r, whose input
-        // windows never touch the halo — launches on the compute stream
-        // concurrently with the exchange on the comm stream; the four
-        // border strips of region(0) launch after the exchange event.
+        // with interior-tile compute on a second HIP stream"): the
+        // INTERIORS of iterations 0..m-1 — regions shrunk so their input
+        // windows never touch exchanged data — launch on the compute
+        // stream concurrently with the exchange on the comm stream;
+        // each iteration's four border BANDS replay afterwards against
+        // that iteration's historical buffer pair (band j reads band
+        // j-1's output and interior j's ring — both already ordered on
+        // the compute stream, so the schedule is race-free by stream
+        // order alone).
+        const int m = std::min(overlapDepth_, T);
         HIP_CHECK(hipEventRecord(evCompute_, compute_));
         HIP_CHECK(hipStreamWaitEvent(comm_, evCompute_, 0));
-        for (auto& b : blocks_) launch_region(b, interior_region(b.g));
+        // exchange enqueues FIRST (host order) and operates on the
+        // chunk-start buffers; it only STARTS after evCompute_, so on
+        // the GPU it runs concurrently with the interiors below
         if (comm_nccl_) {
           exchange_nccl(d);
         } else {
           exchange_local(d);
         }
+        hist_.clear();
+        for (int j = 0; j < m; ++j) {
+          for (auto& b : blocks_)
+            launch_region(b, interior_region(b.g, (j + 1) * r));
+          for (auto& b : blocks_) {
+            hist_.push_back({b.src, b.dst});
+            std::swap(b.src, b.dst);
+          }
+          swapped_ = !swapped_;
+        }
         if (phaseTiming_) HIP_CHECK(hipEventRecord(evPhase_[1], comm_));
         HIP_CHECK(hipEventRecord(evComm_, comm_));
         HIP_CHECK(hipStreamWaitEvent(compute_, evComm_, 0));
-        const int ext0 = d - r;
-        for (auto& b : blocks_) {
-          Region full = region_for(b.g, ext0);
-          Region in = interior_region(b.g);
-          // N / S full-width strips, W / E strips between them
-          Region strips[4] = {
-              {full.py0, in.py0, full.ex0, full.ex1},
-              {in.py1, full.py1, full.ex0, full.ex1},
-              {in.py0, in.py1, full.ex0, in.ex0},
-              {in.py0, in.py1, in.ex1, full.ex1},
-          };
-          for (const Region& s : strips) launch_region(b, s);
+        for (int j = 0; j < m; ++j) {
+          const int ext = d - (j + 1) * r;
+          for (size_t bi = 0; bi < blocks_.size(); ++bi) {
+            auto& b = blocks_[bi];
+            const auto& hp = hist_[(size_t)j * blocks_.size() + bi];
+            Region full = region_for(b.g, ext);
+            Region in = interior_region(b.g, (j + 1) * r);
+            // N / S full-width strips, W / E strips between them
+            Region strips[4] = {
+                {full.py0, in.py0, full.ex0, full.ex1},
+                {in.py1, full.py1, full.ex0, full.ex1},
+                {in.py0, in.py1, full.ex0, in.ex0},
+                {in.py0, in.py1, in.ex1, full.ex1},
+            };
+            for (const Region& s : strips)
+              launch_one(b.g, hp.first, hp.second, s);
+          }
         }
-        for (auto& b : blocks_) std::swap(b.src, b.dst);
-        swapped_ = !swapped_;
-        t0 = 1;
+        t0 = m;
       }
       if (phaseTiming_ && t0 == 0)
         HIP_CHECK(hipEventRecord(evPhase_[1], comm_));  // no exchange
@@ -298,29 +328,50 @@ class ConvEngine {
     return false;
   }
 
-  // iteration-0 interior: the owned block shrunk by r on every side
-  // (empty if the block is thinner than 2r+1 — the border strips then
-  // cover everything)
-  Region interior_region(const BlockGeom& g) const {
-    const int r = k_ / 2;
+  // iteration-j interior: the owned block shrunk by `depth` on every
+  // side — its input windows never touch data the in-flight exchange
+  // will deliver (empty if the block is too thin — the border bands
+  // then cover everything)
+  Region interior_region(const BlockGeom& g, int depth) const {
     Region rg;
-    rg.py0 = g.haloPx + r;
-    rg.py1 = std::max(g.haloPx + g.h - r, rg.py0);
-    rg.ex0 = (g.haloPx + r) * g.S;
-    rg.ex1 = std::max((g.haloPx + g.wPx - r) * g.S, rg.ex0);
+    rg.py0 = g.haloPx + depth;
+    rg.py1 = std::max(g.haloPx + g.h - depth, rg.py0);
+    rg.ex0 = (g.haloPx + depth) * g.S;
+    rg.ex1 = std::max((g.haloPx + g.wPx - depth) * g.S, rg.ex0);
     return rg;
   }
 
-  // single-iteration launch: integer-dot u8 fast path when eligible,
-  // else MFMA/VALU via launch_conv
-  void launch_region(DeviceBlock& b, const Region& rg) {
+  // kernel-path selection on explicit buffers: integer separable / dot
+  // fast paths when eligible, else MFMA/VALU via launch_conv
+  void launch_path(const BlockGeom& g, void* src, void* dst,
+                   const Region& rg) {
     if (sepOn_) {
-      launch_conv_u8sep(b.src, b.dst, sepW_, k_, b.g, rg, compute_);
+      launch_conv_u8sep(src, dst, sepW_, k_, g, rg, compute_);
     } else if (dotShift_ > 0) {
-      launch_conv_u8dot(b.src, b.dst, dotW_, dotShift_, b.g, rg, compute_);
+      launch_conv_u8dot(src, dst, dotW_, dotShift_, g, rg, compute_);
     } else {
-      launch_conv(dt_, S_, k_, b.src, b.dst, (const float*)wgt_, b.g, rg,
-                  useMfma_, compute_);
+      launch_conv(dt_, S_, k_, src, dst, (const float*)wgt_, g, rg, useMfma_,
+                  compute_);
+    }
+  }
+
+  void launch_region(DeviceBlock& b, const Region& rg) {
+    launch_path(b.g, b.src, b.dst, rg);
+  }
+
+  // border-band launch: thin strips route to the element-indexed
+  // generic kernel — the 1024-element tiled kernels waste ~30x on
+  // masked lanes there (bit-compatible: fp32 fma in the same tap order
+  // with the same store rounding; exact classes stay exact)
+  void launch_one(const BlockGeom& g, void* src, void* dst,
+                  const Region& rg) {
+    if (rg.py1 <= rg.py0 || rg.ex1 <= rg.ex0) return;
+    const bool thin = (rg.ex1 - rg.ex0) < 256 || (rg.py1 - rg.py0) < 16;
+    if (thin) {
+      launch_conv_strip(dt_, S_, k_, src, dst, (const float*)wgt_, g, rg,
+                        compute_);
+    } else {
+      launch_path(g, src, dst, rg);
     }
   }
 
@@ -468,6 +519,8 @@ class ConvEngine {
   int dotShift_ = 0;
   bool dotFuse_ = false;
   unsigned dotW_[9] = {};
+  int overlapDepth_ = 2;
+  std::vector<std::pair<void*, void*>> hist_;  // per-overlap-iter buffers
   bool phaseTiming_ = false;
   hipEvent_t evPhase_[4] = {};
   double exchangeMs_ = 0.0;

@@ -524,3 +524,34 @@ def test_sep_vs_dot_identical(monkeypatch):
     monkeypatch.setenv("PCONV_DISABLE_SEP", "1")
     out_dot = _engine_mod().run_single_gpu(spec, img)
     np.testing.assert_array_equal(out_sep, out_dot)
+
+
+# ---- exchange-overlap pyramid (round-2 scheduler) --------------------------
+
+
+@pytest.mark.parametrize("depth", [1, 2, 4])
+@pytest.mark.parametrize("p", [2, 8])
+def test_overlap_depth_byte_exact(depth, p, monkeypatch):
+    """The m-deep interior/band overlap schedule must be bit-identical to
+    the oracle for every depth: interiors of iterations 0..m-1 run while
+    the exchange flies, their border bands replay afterwards against the
+    historical buffer pairs."""
+    monkeypatch.setenv("PCONV_OVERLAP_DEPTH", str(depth))
+    spec = ProblemSpec(160, 144, 1, DType.U8, GAUSSIAN_BLUR_3, 21)
+    img = synthetic_image(144, 160, 1, DType.U8, seed=7)
+    out = _engine_mod().run_local_blocks_gpu(spec, img, p)
+    ref = run_reference(spec, img)
+    np.testing.assert_array_equal(out, ref)
+
+
+def test_overlap_depth_fp16_bit_reproducible(monkeypatch):
+    """f16 storage: multi-block with bands must match the 1-block run
+    bitwise (the generic strip kernel rounds via the same packed cvt)."""
+    spec = ProblemSpec(256, 192, 1, DType.F16, GAUSSIAN_BLUR_3, 12)
+    img = synthetic_image(192, 256, 1, DType.F16, seed=9)
+    ref1 = _engine_mod().run_single_gpu(spec, img)
+    monkeypatch.setenv("PCONV_OVERLAP_DEPTH", "3")
+    out = _engine_mod().run_local_blocks_gpu(spec, img, 4)
+    np.testing.assert_array_equal(
+        out.view(np.uint16), ref1.view(np.uint16)
+    )
