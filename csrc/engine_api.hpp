@@ -8,8 +8,14 @@
 //    kernel launches over shrinking regions — amortizes RCCL latency,
 //    which on xGMI dominates the KB-scale halo volume.
 //  * Two-phase corner-free exchange (parallel/decomp.py docstring):
-//    phase A = E/W column strips (strided → staged via hipMemcpy2DAsync),
-//    phase B = N/S row strips at full pitch (contiguous, sent in place).
+//    phase A = E/W column strips (strided → staged via pack kernels,
+//    launch_copy2d — the SDMA path measured ~8x slower), phase B = N/S
+//    row strips at full pitch (contiguous, sent in place).
+//  * Exchange-overlap pyramid: the INTERIORS of the chunk's first m
+//    iterations run on the compute stream while the exchange flies on
+//    the comm stream; their border bands replay afterwards against the
+//    historical buffer pairs (m = PCONV_OVERLAP_DEPTH, default 2 —
+//    sized for an exchange roughly two interior-iterations long).
 //  * Everything is enqueued asynchronously on two HIP streams (compute +
 //    comm) with event ordering; run() performs ZERO host synchronization
 //    until the end of the whole iteration loop.
