@@ -84,6 +84,18 @@ def main(argv=None) -> int:
     return run_rank(args)
 
 
+def next_stop(done: int, iters: int, converge_every: int, checkpoint_every: int) -> int:
+    """First iteration count > done at which a converge/checkpoint
+    boundary (or the end) falls. Segments run to the NEXT upcoming
+    boundary, so non-divisible --converge-every/--checkpoint-every
+    combinations hit every point."""
+    stops = [iters]
+    for every in (converge_every, checkpoint_every):
+        if every and every > 0:
+            stops.append((done // every + 1) * every)
+    return min(min(stops), iters)
+
+
 def _free_port() -> int:
     import socket
 
@@ -199,17 +211,7 @@ def run_rank(args) -> int:
         def store():
             return blk.store()
 
-    # --- iterate with convergence/checkpoint segments ------------------
-    # Each segment runs to the NEXT upcoming converge/checkpoint boundary
-    # (not a fixed min interval): --converge-every 2 --checkpoint-every 3
-    # stops at 2,3,4,6,... so no boundary is ever skipped.
-    def next_stop(done: int) -> int:
-        stops = [args.iters]
-        for every in (args.converge_every, args.checkpoint_every):
-            if every and every > 0:
-                stops.append((done // every + 1) * every)
-        return min(min(stops), args.iters)
-
+    # --- iterate with convergence/checkpoint segments (next_stop) ------
     if world > 1:
         dist.barrier()
     if backend == "hip":
@@ -217,7 +219,8 @@ def run_rank(args) -> int:
     t0 = time.perf_counter()
     converged = False
     while iters_done < args.iters and not converged:
-        n = next_stop(iters_done) - iters_done
+        n = next_stop(iters_done, args.iters, args.converge_every,
+                      args.checkpoint_every) - iters_done
         if args.watchdog > 0:
             signal.signal(signal.SIGALRM, _alarm_handler)
             signal.alarm(args.watchdog)

@@ -183,3 +183,25 @@ def test_cli_watchdog_smoke(image_and_filter, tmp_path):
     r = run_cli([ip, "64", "48", fp, "3", "--backend", "cpu",
                  "--watchdog", "120", "--no-output"])
     assert r.returncode == 0, r.stderr
+
+
+def test_next_stop_boundaries():
+    """Segments run to the NEXT boundary: --converge-every 2
+    --checkpoint-every 3 over 10 iters stops at 2,3,4,6,8,9,10 — no
+    boundary skipped (ADVICE.md r01 finding)."""
+    from parallel_convolution_amd.cli import next_stop
+
+    stops = []
+    done = 0
+    while done < 10:
+        done = next_stop(done, 10, 2, 3)
+        stops.append(done)
+    assert stops == [2, 3, 4, 6, 8, 9, 10]
+    # single boundary kind
+    assert next_stop(0, 100, 0, 30) == 30
+    assert next_stop(30, 100, 0, 30) == 60
+    assert next_stop(95, 100, 0, 30) == 100
+    # no boundaries: run to the end
+    assert next_stop(0, 7, 0, 0) == 7
+    # boundary beyond the end clamps
+    assert next_stop(0, 5, 8, 0) == 5
