@@ -372,7 +372,12 @@ class ConvEngine {
   void launch_one(const BlockGeom& g, void* src, void* dst,
                   const Region& rg) {
     if (rg.py1 <= rg.py0 || rg.ex1 <= rg.ex0) return;
-    const bool thin = (rg.ex1 - rg.ex0) < 256 || (rg.py1 - rg.py0) < 16;
+    static const bool noStrip = [] {
+      const char* e = getenv("PCONV_STRIP_ROUTE");
+      return e && atoi(e) == 0;
+    }();
+    const bool thin =
+        !noStrip && ((rg.ex1 - rg.ex0) < 256 || (rg.py1 - rg.py0) < 16);
     if (thin) {
       launch_conv_strip(dt_, S_, k_, src, dst, (const float*)wgt_, g, rg,
                         compute_);
