@@ -61,6 +61,12 @@ struct DeviceBlock {
   void* sendE = nullptr;
   void* recvW = nullptr;
   void* recvE = nullptr;
+  // send staging for phase-B row strips (N/S), haloPx * pitchE elements
+  // each: sends must be CAPTURED before the overlap pyramid's interior
+  // j>=1 overwrites the source buffer's edge strips (write-after-read
+  // race, caught by the r02 fuzz campaign at overlap depth 2)
+  void* sendN = nullptr;
+  void* sendS = nullptr;
 };
 
 class ConvEngine {
@@ -84,6 +90,7 @@ class ConvEngine {
     HIP_CHECK(hipStreamCreateWithFlags(&comm_, hipStreamNonBlocking));
     HIP_CHECK(hipEventCreateWithFlags(&evComm_, hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&evCompute_, hipEventDisableTiming));
+    HIP_CHECK(hipEventCreateWithFlags(&evPacked_, hipEventDisableTiming));
     for (int i = 0; i < 4; ++i) HIP_CHECK(hipEventCreate(&evPhase_[i]));
     HIP_CHECK(hipMalloc(&wgt_, filter.size() * sizeof(float)));
     HIP_CHECK(hipMemcpy(wgt_, filter.data(), filter.size() * sizeof(float),
@@ -116,12 +123,15 @@ class ConvEngine {
       if (b.sendE) (void)hipFree(b.sendE);
       if (b.recvW) (void)hipFree(b.recvW);
       if (b.recvE) (void)hipFree(b.recvE);
+      if (b.sendN) (void)hipFree(b.sendN);
+      if (b.sendS) (void)hipFree(b.sendS);
     }
     if (wgt_) (void)hipFree(wgt_);
     if (diffDev_) (void)hipFree(diffDev_);
     if (comm_nccl_) (void)ncclCommDestroy(comm_nccl_);
     (void)hipEventDestroy(evComm_);
     (void)hipEventDestroy(evCompute_);
+    (void)hipEventDestroy(evPacked_);
     for (int i = 0; i < 4; ++i) (void)hipEventDestroy(evPhase_[i]);
     (void)hipStreamDestroy(compute_);
     (void)hipStreamDestroy(comm_);
@@ -133,6 +143,7 @@ class ConvEngine {
     b.src = reinterpret_cast<void*>(src);
     b.dst = reinterpret_cast<void*>(dst);
     const size_t strip = (size_t)g.h * g.haloPx * g.S * itemsize(dt_);
+    const size_t rowsB = (size_t)g.haloPx * g.pitchE * itemsize(dt_);
     if (g.nbr[NB_W] >= 0) {
       HIP_CHECK(hipMalloc(&b.sendW, strip));
       HIP_CHECK(hipMalloc(&b.recvW, strip));
@@ -141,6 +152,8 @@ class ConvEngine {
       HIP_CHECK(hipMalloc(&b.sendE, strip));
       HIP_CHECK(hipMalloc(&b.recvE, strip));
     }
+    if (g.nbr[NB_N] >= 0) HIP_CHECK(hipMalloc(&b.sendN, rowsB));
+    if (g.nbr[NB_S] >= 0) HIP_CHECK(hipMalloc(&b.sendS, rowsB));
     blocks_.push_back(b);
     // fused integer-dot pairs (conv_u8dot2_k) halve HBM traffic but the
     // redundant iteration-A apron costs more than the bandwidth saved —
@@ -205,6 +218,12 @@ class ConvEngine {
         }
         hist_.clear();
         for (int j = 0; j < m; ++j) {
+          if (j == 1) {
+            // interiors j>=1 write buffers the exchange's capture phase
+            // reads (the double-buffer flip aliases them onto the
+            // chunk-start src): wait until every send has left src
+            HIP_CHECK(hipStreamWaitEvent(compute_, evPacked_, 0));
+          }
           for (auto& b : blocks_)
             launch_region(b, interior_region(b.g, (j + 1) * r));
           for (auto& b : blocks_) {
@@ -456,9 +475,22 @@ class ConvEngine {
                                elem(s, s.src, H, 0), rowB * d,
                                hipMemcpyDeviceToDevice, comm_));
     }
+    // all reads of src complete at this point in comm-stream order: the
+    // overlap pyramid's interiors j>=1 (which overwrite src edge
+    // strips) gate on this event
+    HIP_CHECK(hipEventRecord(evPacked_, comm_));
   }
 
   // RCCL mode: one block, neighbors are global ranks.
+  //
+  // All send data leaves `src` in the capture phase below, BEFORE any
+  // network round, so evPacked_ lets the overlap pyramid's interiors
+  // j>=1 overwrite src edge strips while the exchange is in flight
+  // (without staging, iteration 1's interior raced the sends —
+  // caught by the r02 fuzz campaign). The N/S rows' E/W-halo corner
+  // columns only become valid after phase A; they are patched into the
+  // staged rows FROM THE RECV STAGING (never from src) — identical
+  // bytes to what phase A unpacks into src.
   void exchange_nccl(int d) {
     DeviceBlock& b = blocks_[0];
     const int is = itemsize(dt_);
@@ -466,14 +498,24 @@ class ConvEngine {
     const size_t stripB = (size_t)b.g.h * d * b.g.S * is;  // phase-A message
     const size_t wB = (size_t)d * b.g.S * is;              // strip row bytes
     const size_t bp = (size_t)b.g.pitchE * is;
+    const size_t rowsB = (size_t)d * bp;
     const int W = b.g.nbr[NB_W], E = b.g.nbr[NB_E];
     const int N = b.g.nbr[NB_N], Sn = b.g.nbr[NB_S];
 
-    // Phase A: pack E/W strips, grouped send/recv, unpack.
+    // capture phase: E/W column strips and N/S row strips -> staging
     if (W >= 0)
       copy2d(b.sendW, wB, elem(b, b.src, H, H), bp, wB, b.g.h);
     if (E >= 0)
       copy2d(b.sendE, wB, elem(b, b.src, H, H + b.g.wPx - d), bp, wB, b.g.h);
+    if (N >= 0)
+      HIP_CHECK(hipMemcpyAsync(b.sendN, elem(b, b.src, H, 0), rowsB,
+                               hipMemcpyDeviceToDevice, comm_));
+    if (Sn >= 0)
+      HIP_CHECK(hipMemcpyAsync(b.sendS, elem(b, b.src, H + b.g.h - d, 0),
+                               rowsB, hipMemcpyDeviceToDevice, comm_));
+    HIP_CHECK(hipEventRecord(evPacked_, comm_));
+
+    // Phase A: E/W grouped send/recv, unpack into src halo columns.
     if (W >= 0 || E >= 0) {
       NCCL_CHECK(ncclGroupStart());
       if (W >= 0) {
@@ -491,19 +533,30 @@ class ConvEngine {
         copy2d(elem(b, b.src, H, H + b.g.wPx), bp, b.recvE, wB, wB, b.g.h);
     }
 
-    // Phase B: N/S full-pitch rows, contiguous — no staging.
+    // corner patch: the W/E d-wide column ranges of the staged N/S rows
+    // get the just-received phase-A data (corner propagation of the
+    // two-phase scheme), sourced from recvW/recvE staging.
+    auto patch = [&](void* sendRows, size_t recvRowOff) {
+      if (W >= 0)
+        copy2d((char*)sendRows + (size_t)(H - d) * b.g.S * is, bp,
+               (const char*)b.recvW + recvRowOff * wB, wB, wB, d);
+      if (E >= 0)
+        copy2d((char*)sendRows + (size_t)(H + b.g.wPx) * b.g.S * is, bp,
+               (const char*)b.recvE + recvRowOff * wB, wB, wB, d);
+    };
+    if (N >= 0) patch(b.sendN, 0);
+    if (Sn >= 0) patch(b.sendS, (size_t)b.g.h - d);
+
+    // Phase B: N/S full-pitch rows — sends from staging, recvs direct.
     if (N >= 0 || Sn >= 0) {
-      const size_t rowsB = (size_t)d * bp;
       NCCL_CHECK(ncclGroupStart());
       if (N >= 0) {
-        NCCL_CHECK(ncclSend(elem(b, b.src, H, 0), rowsB, ncclChar, N,
-                            comm_nccl_, comm_));
+        NCCL_CHECK(ncclSend(b.sendN, rowsB, ncclChar, N, comm_nccl_, comm_));
         NCCL_CHECK(ncclRecv(elem(b, b.src, H - d, 0), rowsB, ncclChar, N,
                             comm_nccl_, comm_));
       }
       if (Sn >= 0) {
-        NCCL_CHECK(ncclSend(elem(b, b.src, H + b.g.h - d, 0), rowsB, ncclChar,
-                            Sn, comm_nccl_, comm_));
+        NCCL_CHECK(ncclSend(b.sendS, rowsB, ncclChar, Sn, comm_nccl_, comm_));
         NCCL_CHECK(ncclRecv(elem(b, b.src, H + b.g.h, 0), rowsB, ncclChar, Sn,
                             comm_nccl_, comm_));
       }
@@ -519,6 +572,7 @@ class ConvEngine {
   hipStream_t comm_ = nullptr;
   hipEvent_t evComm_ = nullptr;
   hipEvent_t evCompute_ = nullptr;
+  hipEvent_t evPacked_ = nullptr;
   void* wgt_ = nullptr;
   void* diffDev_ = nullptr;
   ncclComm_t comm_nccl_ = nullptr;
