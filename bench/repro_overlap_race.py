@@ -1,4 +1,18 @@
+"""Deterministic regression for the overlap-pyramid send-capture race.
+
+The r02 fuzz campaign caught intermittent corruption at overlap depth
+>= 2 on 2D process grids (write-after-read: interior 1 overwrote the
+chunk-start buffer's edge strips while the exchange was still reading
+them for sends). These fixed cases reproduced it reliably before the
+staged-send fix (csrc/engine_api.hpp exchange capture phase +
+evPacked_); they must stay at BAD: 0.
+
+    python bench/repro_overlap_race.py            # default depth (2)
+    PCONV_OVERLAP_DEPTH=4 python bench/repro_overlap_race.py
+"""
+
 import sys
+
 sys.path.insert(0, ".")
 import numpy as np
 from parallel_convolution_amd.hip_engine import run_local_blocks_gpu
