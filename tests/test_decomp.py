@@ -108,3 +108,27 @@ def test_pitch_alignment():
     for b in blocks:
         assert b.buf_w_elems % 64 == 0
         assert b.buf_w_elems >= (b.w + 2 * b.halo) * 3
+
+
+def test_max_chunk_caps_at_block_dims():
+    """Exchange depth d = r*T must fit the sender's OWNED region on every
+    neighbored axis — the staged N/S corner patches index recv buffers at
+    row h-d and require d <= h (csrc exchange_nccl); same for columns."""
+    from parallel_convolution_amd.parallel.decomp import (
+        halo_for_radius,
+        make_blocks,
+        max_chunk_iters,
+    )
+
+    for (H, W, p, r) in [(55, 662, 8, 1), (27, 500, 8, 2), (500, 30, 6, 2),
+                         (64, 64, 4, 1), (40, 40, 8, 2)]:
+        halo = halo_for_radius(r)
+        blocks = make_blocks(H, W, p, 1, halo=halo)
+        T = max_chunk_iters(blocks, r, halo)
+        d = r * T
+        assert d <= halo
+        for b in blocks:
+            if b.north is not None or b.south is not None:
+                assert d <= b.h, (H, W, p, r, b)
+            if b.west is not None or b.east is not None:
+                assert d <= b.w, (H, W, p, r, b)
